@@ -1,0 +1,42 @@
+"""Backend dispatch: CPU (gloo bootstrap plane) vs GPU (native RCCL/HIP).
+
+The reference dispatches per XLA platform with decorator-wrapped lowerings
+(``/root/reference/mpi4jax/_src/decorators.py``).  Here dispatch is simply by
+tensor device: host tensors ride the gloo bootstrap plane, device tensors go
+through the hand-written HIP/RCCL extension.  There is no copy-to-host mode
+and no multi-vendor dispatch — device traffic is always RCCL over xGMI.
+"""
+
+import torch
+
+
+def is_gpu(x: torch.Tensor) -> bool:
+    return x.is_cuda
+
+
+def backend_for(x: torch.Tensor):
+    if x.is_cuda:
+        from . import rccl
+
+        return rccl
+    from . import cpu
+
+    return cpu
+
+
+def has_rccl_support() -> bool:
+    """True when the native HIP/RCCL extension is importable."""
+    try:
+        from . import rccl
+
+        rccl.ext()
+        return True
+    except Exception:
+        return False
+
+
+# capability-probe parity with the reference API
+# (`mpi4jax.has_cuda_support`, utils.py:159-174) — for us "GPU support"
+# means the RCCL extension.
+def has_cuda_support() -> bool:
+    return has_rccl_support()

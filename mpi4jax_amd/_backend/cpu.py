@@ -1,0 +1,231 @@
+"""CPU transport over the gloo process group.
+
+Functional equivalent of the reference's CPU FFI backend
+(``/root/reference/mpi4jax/_src/xla_bridge/mpi_xla_bridge_cpu.cpp``), used
+for host tensors and for GPU-less correctness testing ("a cluster of one" —
+SURVEY.md §4).  All functions take comm-relative ranks and NEVER mutate
+their inputs (the reference's immutability contract).
+"""
+
+import torch
+import torch.distributed as dist
+
+from ..ops.reduce_ops import Op, GLOO_OP_MAP, combine
+from ..utils.status import ANY_SOURCE, ANY_TAG
+from ..utils.logging import debug_timer
+
+# Per-communicator FIFO queues for same-rank send->recv (self-messaging).
+# MPI supports buffered self sends; gloo does not, so we emulate locally.
+_SELF_QUEUES = {}
+
+
+def _self_queue(comm):
+    return _SELF_QUEUES.setdefault(id(comm), [])
+
+
+def _gloo_reduce_op(op: Op, dtype):
+    if op is Op.AVG:
+        # gloo has no AVG; emulate with SUM + divide
+        return None
+    if op not in GLOO_OP_MAP:
+        raise ValueError(f"unsupported reduction {op} on the CPU backend")
+    return GLOO_OP_MAP[op]
+
+
+def allreduce(x, op, comm):
+    with debug_timer("Allreduce", comm.rank, f"{x.numel()} items"):
+        out = x.clone().contiguous()
+        if comm.size == 1:
+            if op is Op.AVG:
+                return out
+            return out
+        gop = _gloo_reduce_op(op, x.dtype)
+        if gop is None:  # AVG
+            dist.all_reduce(out, op=dist.ReduceOp.SUM, group=comm.gloo_group)
+            out = out / comm.size
+        else:
+            dist.all_reduce(out, op=gop, group=comm.gloo_group)
+        return out
+
+
+def reduce(x, op, root, comm):
+    with debug_timer("Reduce", comm.rank, f"{x.numel()} items"):
+        out = x.clone().contiguous()
+        if comm.size == 1:
+            return out if comm.rank == root else None
+        gop = _gloo_reduce_op(op, x.dtype)
+        if gop is None:
+            dist.reduce(
+                out, dst=comm.global_rank(root), op=dist.ReduceOp.SUM,
+                group=comm.gloo_group,
+            )
+            if comm.rank == root:
+                out = out / comm.size
+        else:
+            dist.reduce(
+                out, dst=comm.global_rank(root), op=gop, group=comm.gloo_group
+            )
+        return out if comm.rank == root else None
+
+
+def allgather(x, comm):
+    with debug_timer("Allgather", comm.rank, f"{x.numel()} items"):
+        xc = x.contiguous()
+        if comm.size == 1:
+            return xc[None].clone()
+        out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
+        dist.all_gather(list(out.unbind(0)), xc, group=comm.gloo_group)
+        return out
+
+
+def alltoall(x, comm):
+    with debug_timer("Alltoall", comm.rank, f"{x.numel()} items"):
+        xc = x.contiguous()
+        if comm.size == 1:
+            return xc.clone()
+        out = torch.empty_like(xc)
+        dist.all_to_all_single(out, xc, group=comm.gloo_group)
+        return out
+
+
+def barrier(comm):
+    with debug_timer("Barrier", comm.rank):
+        if comm.size > 1:
+            dist.barrier(group=comm.gloo_group)
+
+
+def bcast(x, root, comm):
+    with debug_timer("Bcast", comm.rank, f"{x.numel()} items"):
+        out = x.clone().contiguous()
+        if comm.size > 1:
+            dist.broadcast(out, src=comm.global_rank(root), group=comm.gloo_group)
+        return out
+
+
+def gather(x, root, comm):
+    with debug_timer("Gather", comm.rank, f"{x.numel()} items"):
+        xc = x.contiguous()
+        if comm.size == 1:
+            return xc[None].clone() if comm.rank == root else None
+        if comm.rank == root:
+            out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
+            dist.gather(
+                xc, list(out.unbind(0)), dst=comm.global_rank(root),
+                group=comm.gloo_group,
+            )
+            return out
+        dist.gather(xc, None, dst=comm.global_rank(root), group=comm.gloo_group)
+        return None
+
+
+def scatter(x, root, comm):
+    with debug_timer("Scatter", comm.rank):
+        if comm.size == 1:
+            return x[0].clone().contiguous()
+        if comm.rank == root:
+            xc = x.contiguous()
+            out = torch.empty(tuple(x.shape[1:]), dtype=x.dtype)
+            dist.scatter(
+                out, [t.contiguous() for t in xc.unbind(0)],
+                src=comm.global_rank(root), group=comm.gloo_group,
+            )
+        else:
+            out = torch.empty(tuple(x.shape), dtype=x.dtype)
+            dist.scatter(
+                out, None, src=comm.global_rank(root), group=comm.gloo_group
+            )
+        return out
+
+
+def scan(x, op, comm):
+    """Inclusive prefix reduction: out_r = x_0 ⊕ ... ⊕ x_r.
+
+    Implemented as a rank chain (recv partial from r-1, combine, forward to
+    r+1) — the same dataflow the GPU ring uses with the HIP combine kernel.
+    """
+    with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
+        out = x.clone().contiguous()
+        if comm.size == 1:
+            return out
+        if comm.rank > 0:
+            partial = torch.empty_like(out)
+            dist.recv(partial, src=comm.global_rank(comm.rank - 1),
+                      group=comm.gloo_group, tag=771)
+            out = combine(op, partial, out)
+        if comm.rank < comm.size - 1:
+            dist.send(out.contiguous(), dst=comm.global_rank(comm.rank + 1),
+                      group=comm.gloo_group, tag=771)
+        return out
+
+
+def send(x, dest, tag, comm):
+    with debug_timer("Send", comm.rank, f"to {dest}, tag {tag}"):
+        xc = x.contiguous().clone()
+        if dest == comm.rank:
+            _self_queue(comm).append((tag, xc))
+            return
+        dist.send(xc, dst=comm.global_rank(dest), group=comm.gloo_group,
+                  tag=max(tag, 0))
+
+
+def recv(template, source, tag, comm, status):
+    with debug_timer("Recv", comm.rank, f"from {source}, tag {tag}"):
+        out = torch.empty(
+            tuple(template.shape), dtype=template.dtype
+        )
+        if source == comm.rank:
+            q = _self_queue(comm)
+            for i, (t, buf) in enumerate(q):
+                if tag in (ANY_TAG, t):
+                    q.pop(i)
+                    out.copy_(buf)
+                    _fill_status(status, comm.rank, t, out)
+                    return out
+            raise RuntimeError(
+                "recv from self with no matching buffered send"
+            )
+        src = None if source == ANY_SOURCE else comm.global_rank(source)
+        sender = dist.recv(out, src=src, group=comm.gloo_group,
+                           tag=max(tag, 0))
+        src_comm_rank = (
+            source if source != ANY_SOURCE
+            else comm._ranks.index(sender)
+        )
+        _fill_status(status, src_comm_rank, tag, out)
+        return out
+
+
+def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
+    with debug_timer(
+        "Sendrecv", comm.rank, f"src {source} dst {dest}"
+    ):
+        out = torch.empty(tuple(recvbuf.shape), dtype=recvbuf.dtype)
+        if source == comm.rank and dest == comm.rank:
+            out.copy_(sendbuf.reshape(out.shape))
+            _fill_status(status, source, recvtag, out)
+            return out
+        sc = sendbuf.contiguous()
+        # even ranks send first — deadlock-free pairing for blocking gloo p2p
+        first_send = comm.rank <= source if dest != comm.rank else False
+        ops = []
+        sreq = dist.P2POp(
+            dist.isend, sc, peer=comm.global_rank(dest),
+            group=comm.gloo_group, tag=max(sendtag, 0),
+        )
+        rreq = dist.P2POp(
+            dist.irecv, out, peer=comm.global_rank(source),
+            group=comm.gloo_group, tag=max(recvtag, 0),
+        )
+        ops = [sreq, rreq] if not first_send else [rreq, sreq]
+        reqs = dist.batch_isend_irecv(ops)
+        for r in reqs:
+            r.wait()
+        _fill_status(status, source, recvtag, out)
+        return out
+
+
+def _fill_status(status, source, tag, out):
+    if status is not None:
+        status.source = source
+        status.tag = tag
+        status.count = out.numel() * out.element_size()

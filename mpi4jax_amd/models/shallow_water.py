@@ -1,0 +1,313 @@
+"""Distributed nonlinear shallow-water solver (torch, MI355X-native).
+
+This is the framework's flagship demo/benchmark workload, matching the
+reference demo's physics and decomposition so its published numbers are
+directly comparable (``/root/reference/examples/shallow_water.py``; the
+scheme is the public Sadourny C-grid solver from dionhaefner/shallow-water).
+Domain decomposition is a 2-D :class:`CartesianGrid`; every model step does
+~10 halo exchanges through mpi4jax_amd ``sendrecv``/``send``/``recv``
+(reference call stack: SURVEY.md §3.4).
+
+The reference's benchmark config (docs/shallow-water.rst:49-52) is a
+(3600, 1800) float32 domain run for 0.1 model days; wall-clock per model
+day is the headline metric.
+
+Implementation is eager torch ops; the ``fused`` path (HIP kernels +
+hipGraph capture) plugs in at :meth:`step` without changing semantics.
+"""
+
+import math
+import time
+from collections import namedtuple
+
+import torch
+
+from ..parallel.comm import resolve_comm
+from ..parallel.grid import CartesianGrid, default_dims
+
+ModelState = namedtuple("ModelState", "h u v dh du dv")
+
+DAY_IN_SECONDS = 86_400.0
+GRAVITY = 9.81
+DEPTH = 100.0
+CORIOLIS_F = 2e-4
+CORIOLIS_BETA = 2e-11
+ADAMS_BASHFORTH_A = 1.5 + 0.1
+ADAMS_BASHFORTH_B = -(0.5 + 0.1)
+
+_I = slice(1, -1)  # interior
+_L = slice(None, -2)  # shifted left/down
+_R = slice(2, None)  # shifted right/up
+
+
+class ShallowWater:
+    """Nonlinear shallow-water model on a distributed C-grid.
+
+    Arguments:
+        nx, ny: *global interior* grid size (the demo default is 360×180,
+            benchmark mode 3600×1800).
+        comm: communicator (defaults to the world).
+        dims: process grid (nproc_y, nproc_x); default mirrors the
+            reference demo (min(size,2), size/nproc_y).
+        device / dtype: tensor placement (float32 matches the reference).
+        periodic_x: periodic east-west boundary (reference default True).
+    """
+
+    def __init__(self, nx=360, ny=180, dx=5e3, dy=5e3, *, comm=None,
+                 dims=None, device="cpu", dtype=torch.float32,
+                 periodic_x=True, lateral_viscosity=None):
+        self.comm = resolve_comm(comm)
+        self.grid = CartesianGrid(self.comm, dims=dims,
+                                  periodic=(False, periodic_x))
+        self.device = torch.device(device)
+        self.dtype = dtype
+        self.dx, self.dy = float(dx), float(dy)
+        self.periodic_x = periodic_x
+
+        self.nx_global = nx + 2
+        self.ny_global = ny + 2
+        npy, npx = self.grid.nproc_y, self.grid.nproc_x
+        if nx % npx or ny % npy:
+            raise ValueError(
+                f"domain {ny}x{nx} not divisible by process grid {npy}x{npx}"
+            )
+        self.nx_local = nx // npx + 2
+        self.ny_local = ny // npy + 2
+
+        iy, ix = self.grid.coords
+        # global index of this rank's first (halo) row/col
+        self.y0 = (self.ny_local - 2) * iy
+        self.x0 = (self.nx_local - 2) * ix
+
+        # coordinates (match the reference's layout: index -1..n along each
+        # axis times the spacing, shallow_water.py:86-90)
+        x_g = (torch.arange(-1, self.nx_global - 1, dtype=torch.float64)
+               * self.dx)
+        y_g = (torch.arange(-1, self.ny_global - 1, dtype=torch.float64)
+               * self.dy)
+        self.length_x = float(x_g[-2] - x_g[1])
+        self.length_y = float(y_g[-2] - y_g[1])
+        self.x_local = x_g[self.x0:self.x0 + self.nx_local]
+        self.y_local = y_g[self.y0:self.y0 + self.ny_local]
+
+        yy = self.y_local[:, None].expand(self.ny_local, self.nx_local)
+        self.coriolis = (CORIOLIS_F + yy * CORIOLIS_BETA).to(
+            self.device, self.dtype
+        )
+        if lateral_viscosity is None:
+            lateral_viscosity = 1e-3 * CORIOLIS_F * self.dx ** 2
+        self.lateral_viscosity = float(lateral_viscosity)
+
+        # CFL time step (shallow_water.py:135)
+        self.dt = 0.125 * min(self.dx, self.dy) / math.sqrt(GRAVITY * DEPTH)
+
+    # ------------------------------------------------------------------
+    def _at_north_edge(self):
+        return self.grid.coords[0] == self.grid.nproc_y - 1
+
+    def _at_east_edge(self):
+        return self.grid.coords[1] == self.grid.nproc_x - 1
+
+    def enforce_boundaries(self, arr, kind):
+        """Halo exchange + physical wall conditions for one field."""
+        arr = self.grid.halo_exchange(arr)
+        if not self.periodic_x and kind == "u" and self._at_east_edge():
+            arr[:, -2] = 0.0
+        if kind == "v" and self._at_north_edge():
+            arr[-2, :] = 0.0
+        return arr
+
+    # ------------------------------------------------------------------
+    def initial_conditions(self):
+        """Zonal jet in geostrophic balance + small perturbation."""
+        dev, dt_ = self.device, self.dtype
+        ny_g, nx_g = self.ny_global, self.nx_global
+
+        # u depends only on y -> compute the global 1-D profile, then the
+        # (cheap, rank-local) global cumsum for geostrophic h
+        y_g = (torch.arange(-1, ny_g - 1, dtype=torch.float64) * self.dy)
+        u_prof = 10.0 * torch.exp(
+            -((y_g - 0.5 * self.length_y) ** 2) / (0.02 * self.length_x) ** 2
+        )
+        cor_prof = CORIOLIS_F + y_g * CORIOLIS_BETA
+        h_geo = torch.cumsum(-self.dy * u_prof * cor_prof / GRAVITY, dim=0)
+        h_geo = h_geo - h_geo.mean()
+
+        x_l = self.x_local.to(dev)
+        y_l = self.y_local.to(dev)
+        u0 = (u_prof[self.y0:self.y0 + self.ny_local]
+              .to(dev)[:, None].expand(self.ny_local, self.nx_local))
+        pert = (
+            0.2
+            * torch.sin(x_l / self.length_x * 10 * math.pi)[None, :]
+            * torch.cos(y_l / self.length_y * 8 * math.pi)[:, None]
+        )
+        h0 = (DEPTH
+              + h_geo[self.y0:self.y0 + self.ny_local].to(dev)[:, None]
+              + pert)
+        h0 = h0.to(dt_).contiguous()
+        u0 = u0.to(dt_).contiguous()
+        v0 = torch.zeros_like(u0)
+
+        h0 = self.enforce_boundaries(h0, "h")
+        u0 = self.enforce_boundaries(u0, "u")
+        v0 = self.enforce_boundaries(v0, "v")
+
+        z = torch.zeros_like(h0)
+        return ModelState(h0, u0, v0, z.clone(), z.clone(), z.clone())
+
+    # ------------------------------------------------------------------
+    def _edge_pad_interior(self, h):
+        """hc = interior of h padded by edge replication (halo ring)."""
+        hc = h.clone()
+        hc[0, :] = hc[1, :]
+        hc[-1, :] = hc[-2, :]
+        hc[:, 0] = hc[:, 1]
+        hc[:, -1] = hc[:, -2]
+        return hc
+
+    def step(self, state, first_step=False):
+        """One model step (Euler on the first step, AB2 afterwards)."""
+        h, u, v, dh, du, dv = state
+        dx, dy = self.dx, self.dy
+        eb = self.enforce_boundaries
+
+        hc = self._edge_pad_interior(h)
+        hc = eb(hc, "h")
+
+        # mass fluxes on cell faces
+        fe = torch.zeros_like(u)
+        fn = torch.zeros_like(u)
+        fe[_I, _I] = 0.5 * (hc[_I, _I] + hc[_I, _R]) * u[_I, _I]
+        fn[_I, _I] = 0.5 * (hc[_I, _I] + hc[_R, _I]) * v[_I, _I]
+        fe = eb(fe, "u")
+        fn = eb(fn, "v")
+
+        dh_new = torch.zeros_like(dh)
+        dh_new[_I, _I] = (
+            -(fe[_I, _I] - fe[_I, _L]) / dx - (fn[_I, _I] - fn[_L, _I]) / dy
+        )
+
+        # potential vorticity (planetary + relative, / layer thickness)
+        q = torch.zeros_like(u)
+        q[_I, _I] = self.coriolis[_I, _I] + (
+            (v[_I, _R] - v[_I, _I]) / dx - (u[_R, _I] - u[_I, _I]) / dy
+        )
+        q[_I, _I] *= 1.0 / (
+            0.25 * (hc[_I, _I] + hc[_I, _R] + hc[_R, _I] + hc[_R, _R])
+        )
+        q = eb(q, "h")
+
+        du_new = torch.zeros_like(du)
+        dv_new = torch.zeros_like(dv)
+        du_new[_I, _I] = (
+            -GRAVITY * (h[_I, _R] - h[_I, _I]) / dx
+            + 0.5 * (
+                q[_I, _I] * 0.5 * (fn[_I, _I] + fn[_I, _R])
+                + q[_L, _I] * 0.5 * (fn[_L, _I] + fn[_L, _R])
+            )
+        )
+        dv_new[_I, _I] = (
+            -GRAVITY * (h[_R, _I] - h[_I, _I]) / dy
+            - 0.5 * (
+                q[_I, _I] * 0.5 * (fe[_I, _I] + fe[_R, _I])
+                + q[_I, _L] * 0.5 * (fe[_I, _L] + fe[_R, _L])
+            )
+        )
+
+        # kinetic energy gradient
+        ke = torch.zeros_like(u)
+        ke[_I, _I] = 0.5 * (
+            0.5 * (u[_I, _I] ** 2 + u[_I, _L] ** 2)
+            + 0.5 * (v[_I, _I] ** 2 + v[_L, _I] ** 2)
+        )
+        ke = eb(ke, "h")
+        du_new[_I, _I] -= (ke[_I, _R] - ke[_I, _I]) / dx
+        dv_new[_I, _I] -= (ke[_R, _I] - ke[_I, _I]) / dy
+
+        # time integration
+        h = h.clone()
+        u = u.clone()
+        v = v.clone()
+        dtt = self.dt
+        if first_step:
+            u[_I, _I] += dtt * du_new[_I, _I]
+            v[_I, _I] += dtt * dv_new[_I, _I]
+            h[_I, _I] += dtt * dh_new[_I, _I]
+        else:
+            u[_I, _I] += dtt * (ADAMS_BASHFORTH_A * du_new[_I, _I]
+                                + ADAMS_BASHFORTH_B * du[_I, _I])
+            v[_I, _I] += dtt * (ADAMS_BASHFORTH_A * dv_new[_I, _I]
+                                + ADAMS_BASHFORTH_B * dv[_I, _I])
+            h[_I, _I] += dtt * (ADAMS_BASHFORTH_A * dh_new[_I, _I]
+                                + ADAMS_BASHFORTH_B * dh[_I, _I])
+
+        h = eb(h, "h")
+        u = eb(u, "u")
+        v = eb(v, "v")
+
+        # lateral friction
+        if self.lateral_viscosity > 0:
+            nu = self.lateral_viscosity
+            gu = torch.zeros_like(u)
+            gv = torch.zeros_like(u)
+            gu[_I, _I] = nu * (u[_I, _R] - u[_I, _I]) / dx
+            gv[_I, _I] = nu * (u[_R, _I] - u[_I, _I]) / dy
+            gu = eb(gu, "u")
+            gv = eb(gv, "v")
+            u = u.clone()
+            u[_I, _I] += dtt * ((gu[_I, _I] - gu[_I, _L]) / dx
+                                + (gv[_I, _I] - gv[_L, _I]) / dy)
+            gu = torch.zeros_like(v)
+            gv = torch.zeros_like(v)
+            gu[_I, _I] = nu * (v[_I, _R] - v[_I, _I]) / dx
+            gv[_I, _I] = nu * (v[_R, _I] - v[_I, _I]) / dy
+            gu = eb(gu, "u")
+            gv = eb(gv, "v")
+            v = v.clone()
+            v[_I, _I] += dtt * ((gu[_I, _I] - gu[_I, _L]) / dx
+                                + (gv[_I, _I] - gv[_L, _I]) / dy)
+
+        return ModelState(h, u, v, dh_new, du_new, dv_new)
+
+    # ------------------------------------------------------------------
+    def solve(self, t1_seconds, num_multisteps=100, state=None,
+              collect=False):
+        """Iterate to t1; returns (final_state, steps_taken, wall_seconds)."""
+        if state is None:
+            state = self.initial_conditions()
+        sol = [state] if collect else None
+        state = self.step(state, first_step=True)
+        steps = 1
+        t = self.dt
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        start = time.perf_counter()
+        while t < t1_seconds:
+            for _ in range(num_multisteps):
+                state = self.step(state)
+            steps += num_multisteps
+            t += self.dt * num_multisteps
+            if collect:
+                sol.append(state)
+        if self.device.type == "cuda":
+            torch.cuda.synchronize()
+        wall = time.perf_counter() - start
+        if collect:
+            return state, steps, wall, sol
+        return state, steps, wall
+
+    def steps_per_model_day(self):
+        return DAY_IN_SECONDS / self.dt
+
+    def gather_global(self, field):
+        """Gather a distributed field's interior onto rank 0 (debug/plot)."""
+        from ..ops.gather import gather
+
+        parts = gather(field[_I, _I].contiguous(), root=0, comm=self.comm)
+        if self.comm.rank != 0:
+            return None
+        npy, npx = self.grid.nproc_y, self.grid.nproc_x
+        ny_l, nx_l = self.ny_local - 2, self.nx_local - 2
+        out = parts.reshape(npy, npx, ny_l, nx_l)
+        return out.permute(0, 2, 1, 3).reshape(npy * ny_l, npx * nx_l)

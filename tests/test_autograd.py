@@ -1,0 +1,62 @@
+"""Autodiff rules (single-process leg).
+
+Reference semantics: allreduce(SUM) VJP is the identity and its JVP is an
+allreduce of the tangent (``allreduce.py:138-159``); sendrecv's VJP swaps
+source/dest (``sendrecv.py:278-293``).
+"""
+
+import pytest
+import torch
+import torch.autograd.forward_ad as fwd_ad
+
+import mpi4jax_amd as m
+
+
+def test_allreduce_backward_identity():
+    x = torch.randn(4, requires_grad=True)
+    y = m.allreduce(x, m.SUM)
+    g = torch.randn(4)
+    y.backward(g)
+    assert torch.equal(x.grad, g)
+
+
+def test_allreduce_grad_of_sum():
+    x = torch.randn(5, requires_grad=True)
+    loss = m.allreduce(x, m.SUM).sum()
+    loss.backward()
+    assert torch.equal(x.grad, torch.ones(5))
+
+
+def test_allreduce_jvp():
+    x = torch.randn(4)
+    t = torch.randn(4)
+    with fwd_ad.dual_level():
+        xd = fwd_ad.make_dual(x, t)
+        yd = m.allreduce(xd, m.SUM)
+        y, yt = fwd_ad.unpack_dual(yd)
+    assert torch.allclose(y, x)
+    assert torch.allclose(yt, t)
+
+
+def test_allreduce_nonsum_grad_raises():
+    x = torch.randn(4, requires_grad=True)
+    y = m.allreduce(x, m.MAX)
+    with pytest.raises(RuntimeError, match="only differentiable"):
+        y.sum().backward()
+
+
+def test_sendrecv_backward_self():
+    x = torch.randn(4, requires_grad=True)
+    y = m.sendrecv(x, x.detach(), source=0, dest=0)
+    g = torch.randn(4)
+    y.backward(g)
+    assert torch.equal(x.grad, g)
+
+
+def test_chained_ops_grad():
+    # distributed matvec pattern (reference test_allreduce_matvec.py)
+    A = torch.randn(3, 3)
+    x = torch.randn(3, requires_grad=True)
+    y = m.allreduce(A @ x, m.SUM)
+    y.sum().backward()
+    assert torch.allclose(x.grad, A.t() @ torch.ones(3))

@@ -1,0 +1,194 @@
+"""Multi-process (gloo, world_size=2) semantics of all 12 ops.
+
+The distributed path that runs on MI355X via RCCL shares all op-layer code
+with this CPU path; only the transport differs, so these tests pin the
+rank/shape/semantics contracts for both.
+"""
+
+import pytest
+import torch
+
+import mpi4jax_amd as m
+from tests._mp import run_multiproc
+
+
+def _all_collectives(rank, ws):
+    x = torch.arange(6, dtype=torch.float32).reshape(2, 3) + rank
+    ref = x.clone()
+
+    # allreduce: sum over ranks
+    y = m.allreduce(x, m.SUM)
+    expect = sum(
+        torch.arange(6, dtype=torch.float32).reshape(2, 3) + r
+        for r in range(ws)
+    )
+    assert torch.equal(y, expect), (rank, "allreduce")
+
+    assert torch.equal(m.allreduce(x, m.MAX),
+                       torch.arange(6.).reshape(2, 3) + ws - 1)
+    assert torch.allclose(m.allreduce(x, m.AVG), expect / ws)
+
+    # allgather
+    y = m.allgather(x)
+    assert y.shape == (ws, 2, 3)
+    for r in range(ws):
+        assert torch.equal(y[r], torch.arange(6.).reshape(2, 3) + r)
+
+    # alltoall: row i of rank r is r*10+i; after, row i of rank r is i*10+r
+    a = torch.stack([torch.full((3,), float(rank * 10 + i))
+                     for i in range(ws)])
+    y = m.alltoall(a)
+    for i in range(ws):
+        assert torch.equal(y[i], torch.full((3,), float(i * 10 + rank)))
+
+    # bcast
+    y = m.bcast(x, 1)
+    if rank == 1:
+        assert y is x
+    else:
+        assert torch.equal(y, torch.arange(6.).reshape(2, 3) + 1)
+
+    # gather
+    y = m.gather(x, 0)
+    if rank == 0:
+        assert y.shape == (ws, 2, 3)
+        for r in range(ws):
+            assert torch.equal(y[r], torch.arange(6.).reshape(2, 3) + r)
+    else:
+        assert y is x  # non-root passthrough (gather.py:140-150)
+
+    # scatter
+    if rank == 0:
+        src = torch.stack([torch.full((2, 3), float(r)) for r in range(ws)])
+        y = m.scatter(src, 0)
+    else:
+        y = m.scatter(torch.empty(2, 3), 0)
+    assert torch.equal(y, torch.full((2, 3), float(rank)))
+
+    # reduce
+    y = m.reduce(x, m.SUM, 0)
+    if rank == 0:
+        assert torch.equal(y, expect)
+    else:
+        assert y is x
+
+    # scan (inclusive prefix)
+    y = m.scan(x, m.SUM)
+    expect_scan = sum(
+        torch.arange(6.).reshape(2, 3) + r for r in range(rank + 1)
+    )
+    assert torch.equal(y, expect_scan)
+
+    # send/recv pair
+    status = m.Status()
+    if rank == 0:
+        m.send(x, 1, tag=3)
+        got = m.recv(x, 1, tag=4, status=status)
+        assert torch.equal(got, torch.arange(6.).reshape(2, 3) + 1)
+        assert status.source == 1 and status.tag == 4
+    else:
+        got = m.recv(x, 0, tag=3)
+        assert torch.equal(got, torch.arange(6.).reshape(2, 3))
+        m.send(x, 0, tag=4)
+
+    # sendrecv ring
+    other = (rank + 1) % ws
+    y = m.sendrecv(x, x, source=other, dest=other)
+    assert torch.equal(y, torch.arange(6.).reshape(2, 3) + other)
+
+    # input never mutated by any of the above
+    assert torch.equal(x, ref)
+
+    m.barrier()
+
+
+def test_all_collectives():
+    run_multiproc(_all_collectives, 2)
+
+
+def _autograd(rank, ws):
+    x = (torch.arange(4.0) + rank).requires_grad_()
+    y = m.allreduce(x, m.SUM)
+    g = torch.full((4,), float(rank + 1))
+    y.backward(g)
+    # VJP of allreduce(SUM) is the identity (per-rank cotangent unchanged)
+    assert torch.equal(x.grad, g)
+
+    # sendrecv grad routes the cotangent back along the reversed edge
+    other = (rank + 1) % ws
+    x2 = (torch.arange(3.0) + rank).requires_grad_()
+    y2 = m.sendrecv(x2, x2.detach(), source=other, dest=other)
+    assert torch.equal(y2, torch.arange(3.0) + other)
+    y2.backward(torch.full((3,), float(rank)))
+    # rank r receives the cotangent produced on rank `other`
+    assert torch.equal(x2.grad, torch.full((3,), float(other)))
+
+
+def test_autograd_multiproc():
+    run_multiproc(_autograd, 2)
+
+
+def _comm_management(rank, ws):
+    world = m.get_world()
+    assert world.Get_size() == ws and world.Get_rank() == rank
+
+    clone = world.Clone()
+    assert clone.Get_size() == ws
+    x = torch.tensor([float(rank + 1)])
+    assert m.allreduce(x, m.SUM, comm=clone).item() == sum(
+        r + 1 for r in range(ws)
+    )
+
+    # Split into singleton comms by color=rank
+    sub = world.Split(color=rank, key=0)
+    assert sub.Get_size() == 1 and sub.Get_rank() == 0
+    assert m.allreduce(x, m.SUM, comm=sub).item() == rank + 1
+
+    # Split into one group
+    sub2 = world.Split(color=0, key=-rank)  # reversed order
+    assert sub2.Get_size() == ws
+    assert sub2.Get_rank() == ws - 1 - rank
+
+    # default comm is created lazily and cached
+    d1 = m.get_default_comm()
+    d2 = m.get_default_comm()
+    assert d1 is d2
+
+
+def test_comm_management():
+    run_multiproc(_comm_management, 2)
+
+
+def _barrier_ordering(rank, ws, path):
+    import time
+
+    if rank == 1:
+        time.sleep(0.5)
+    with open(f"{path}/r{rank}.start", "w") as f:
+        f.write("s")
+    m.barrier()
+    # after the barrier every rank's start file must exist
+    import os
+
+    for r in range(ws):
+        assert os.path.exists(f"{path}/r{r}.start"), (rank, r)
+
+
+def test_barrier_ordering(tmp_path):
+    run_multiproc(_barrier_ordering, 2, args=(str(tmp_path),))
+
+
+def _world4(rank, ws):
+    x = torch.tensor([1.0 + rank])
+    assert m.allreduce(x, m.SUM).item() == sum(1.0 + r for r in range(ws))
+    y = m.scan(x, m.SUM)
+    assert y.item() == sum(1.0 + r for r in range(rank + 1))
+    a = torch.arange(float(ws)) + rank * 100
+    z = m.alltoall(a[:, None])
+    for i in range(ws):
+        assert z[i, 0].item() == rank + i * 100
+
+
+@pytest.mark.slow
+def test_world4():
+    run_multiproc(_world4, 4)
