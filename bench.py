@@ -1,0 +1,137 @@
+"""Flagship benchmark: shallow-water solver steps/sec on MI355X.
+
+Measures the reference's headline benchmark (BASELINE.json / BASELINE.md):
+the shallow-water demo on a (3600, 1800) float32 domain
+(``/root/reference/docs/shallow-water.rst:49-52``), reported as whole-job
+steps/sec (strong scaling: fixed global domain decomposed over N GPUs), plus
+the in-stream 256 MiB bf16 allreduce bandwidth when N > 1.
+
+Launch (driver contract):
+    python bench.py --gpus N --steps K --warmup W
+N>1 runs under torch.distributed.run with one rank per GPU over RCCL.
+
+vs_baseline compares seconds-per-model-day against the reference's
+published GPU n=1 number (103.18 s/model-day on a Tesla P100,
+docs/shallow-water.rst:81-83).
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+import mpi4jax_amd as m
+from mpi4jax_amd.models import ShallowWater
+from mpi4jax_amd.parallel.grid import default_dims
+
+REF_SEC_PER_MODEL_DAY_GPU1 = 103.18  # P100, docs/shallow-water.rst:81-83
+
+
+def measure_allreduce_gbps(steps=20, warmup=5):
+    """In-stream allreduce bandwidth, 256 MiB bf16 (BASELINE config #2)."""
+    comm = m.get_world()
+    nbytes = 256 * 1024 * 1024
+    n = nbytes // 2
+    x = torch.randn(n, device="cuda").to(torch.bfloat16)
+    for _ in range(warmup):
+        x = m.allreduce(x, m.SUM)
+    m.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        y = m.allreduce(x, m.SUM)
+    m.barrier()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / steps
+    size = comm.size
+    algbw = nbytes / dt / 1e9
+    busbw = algbw * (2 * (size - 1) / size) if size > 1 else algbw
+    return {"algbw_GBps": round(algbw, 2), "busbw_GBps": round(busbw, 2),
+            "bytes": nbytes, "dtype": "bf16"}
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=100)
+    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--nx", type=int, default=3600)
+    p.add_argument("--ny", type=int, default=1800)
+    p.add_argument("--allreduce-bench", action="store_true", default=True)
+    args = p.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    m.init()
+    comm = m.get_world()
+    rank = comm.rank
+    n = comm.size
+    use_gpu = torch.cuda.is_available()
+    device = "cuda" if use_gpu else "cpu"
+
+    dims = default_dims(n)
+    sw = ShallowWater(nx=args.nx, ny=args.ny, comm=comm, dims=dims,
+                      device=device, dtype=torch.float32)
+    state = sw.initial_conditions()
+    state = sw.step(state, first_step=True)
+
+    # warmup (untimed)
+    for _ in range(args.warmup):
+        state = sw.step(state)
+
+    m.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        state = sw.step(state)
+    m.barrier()
+    if use_gpu:
+        torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (job time = slowest rank)
+    elapsed_t = torch.tensor([elapsed], dtype=torch.float64)
+    elapsed = m.allreduce(elapsed_t, m.MAX, comm=comm).item()
+
+    # sanity: the state must still be finite (no skipped work)
+    assert torch.isfinite(state.h).all(), "model diverged"
+
+    steps_per_sec = args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000
+    spd = sw.steps_per_model_day()
+    sec_per_model_day = spd / steps_per_sec
+
+    ar = None
+    if use_gpu:
+        ar = measure_allreduce_gbps()
+
+    if rank == 0:
+        result = {
+            "metric": "shallow_water_steps_per_sec",
+            "value": round(steps_per_sec, 3),
+            "unit": "steps/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "strong",
+            "vs_baseline": round(REF_SEC_PER_MODEL_DAY_GPU1
+                                 / sec_per_model_day, 3),
+            "dtype": "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": "shallow_water",
+                "domain": [args.ny, args.nx],
+                "sec_per_model_day": round(sec_per_model_day, 2),
+                "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}",
+                "allreduce_256MiB_bf16": ar,
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+
+if __name__ == "__main__":
+    main()

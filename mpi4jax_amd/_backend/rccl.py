@@ -10,9 +10,14 @@ host staging anywhere on this path (the reference's COPY_TO_HOST mode,
 
 On a GPU box this module refuses to fall back: if the extension is missing,
 every op raises ImportError loudly rather than silently using eager torch.
-"""
 
-import os
+Dtype strategy:
+* pure data movement (allgather/alltoall/bcast/gather/scatter/p2p) ships
+  raw bytes — every torch dtype travels;
+* arithmetic reductions use the native RCCL dtype; complex reduces as
+  pairs of reals (SUM/AVG only), bool maps to uint8 with OR/AND semantics,
+  int16 upcasts to int32 on-stream (RCCL has no int16).
+"""
 
 import torch
 
@@ -23,6 +28,11 @@ from ..utils.logging import debug_timer, get_logging
 
 _EXT = None
 _EXT_ERR = None
+
+# per-comm FIFO for unmatched self-sends: an ncclSend to self outside a
+# group would block the stream forever, so self p2p short-circuits to a
+# stream-ordered device copy (MPI buffered-self-send semantics).
+_SELF_QUEUES = {}
 
 
 def ext():
@@ -56,11 +66,9 @@ def _handle(comm):
     return comm.rccl_handle()
 
 
-def _as_real(x):
-    """View complex tensors as real for transport (elementwise-sum safe)."""
-    if x.dtype in COMPLEX_AS_REAL:
-        return torch.view_as_real(x.contiguous())
-    return x
+def _bytes(x):
+    """Byte view of a contiguous tensor (shares storage)."""
+    return x.flatten().view(torch.uint8)
 
 
 def _check_op(op, x, op_name):
@@ -69,37 +77,76 @@ def _check_op(op, x, op_name):
             f"{op_name}: reduction {op} is not supported on the RCCL "
             f"backend (supported: SUM, PROD, MIN, MAX, AVG)"
         )
-    if x.is_complex() and op is not Op.SUM and op is not Op.AVG:
+    if x.is_complex() and op not in (Op.SUM, Op.AVG):
         raise ValueError(f"{op_name}: {op} is undefined for complex dtypes")
 
 
-def _alloc_out(shape, like):
-    return torch.empty(tuple(shape), dtype=like.dtype, device=like.device)
+# bool reductions: OR for SUM/MAX, AND for PROD/MIN (MPI logical semantics)
+_BOOL_OP = {Op.SUM: Op.MAX, Op.MAX: Op.MAX, Op.PROD: Op.MIN, Op.MIN: Op.MIN}
+
+
+def _reduction_view(x, op, op_name):
+    """Map (tensor, op) onto an RCCL-native (tensor, op, postprocess)."""
+    if x.is_complex():
+        return torch.view_as_real(x), op, None
+    if x.dtype == torch.bool:
+        if op not in _BOOL_OP:
+            raise ValueError(f"{op_name}: {op} undefined for bool")
+        return x.to(torch.uint8), _BOOL_OP[op], lambda t: t.to(torch.bool)
+    if x.dtype == torch.int16:
+        return x.to(torch.int32), op, lambda t: t.to(torch.int16)
+    return x, op, None
 
 
 def allreduce(x, op, comm):
     _check_op(op, x, "allreduce")
     with debug_timer("Allreduce", comm.rank, f"{x.numel()} items"):
-        out = torch.empty_like(x, memory_format=torch.contiguous_format)
-        ext().allreduce(_as_real(out), _as_real(x.contiguous()),
-                        RCCL_OP_ENUM[op], _handle(comm))
-        return out
+        xr, rop, post = _reduction_view(x.contiguous(), op, "allreduce")
+        out = torch.empty_like(xr)
+        ext().allreduce(out, xr, RCCL_OP_ENUM[rop], _handle(comm))
+        if post is not None:
+            out = post(out)
+        if x.is_complex():
+            out = torch.view_as_complex(out)
+        return out.reshape(x.shape)
 
 
 def reduce(x, op, root, comm):
     _check_op(op, x, "reduce")
     with debug_timer("Reduce", comm.rank, f"{x.numel()} items"):
-        out = torch.empty_like(x, memory_format=torch.contiguous_format)
-        ext().reduce(_as_real(out), _as_real(x.contiguous()),
-                     RCCL_OP_ENUM[op], root, _handle(comm))
-        return out if comm.rank == root else None
+        xr, rop, post = _reduction_view(x.contiguous(), op, "reduce")
+        out = torch.empty_like(xr)
+        ext().reduce(out, xr, RCCL_OP_ENUM[rop], root, _handle(comm))
+        if comm.rank != root:
+            return None
+        if post is not None:
+            out = post(out)
+        if x.is_complex():
+            out = torch.view_as_complex(out)
+        return out.reshape(x.shape)
+
+
+def scan(x, op, comm):
+    _check_op(op, x, "scan")
+    if op is Op.AVG:
+        raise ValueError("scan: AVG is not a valid scan operator")
+    with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
+        xr, rop, post = _reduction_view(x.contiguous(), op, "scan")
+        out = torch.empty_like(xr)
+        ext().scan(out, xr, RCCL_OP_ENUM[rop], _handle(comm))
+        if post is not None:
+            out = post(out)
+        if x.is_complex():
+            out = torch.view_as_complex(out)
+        return out.reshape(x.shape)
 
 
 def allgather(x, comm):
     with debug_timer("Allgather", comm.rank, f"{x.numel()} items"):
         xc = x.contiguous()
-        out = _alloc_out((comm.size,) + tuple(x.shape), x)
-        ext().allgather(_as_real(out), _as_real(xc), _handle(comm))
+        out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype,
+                          device=x.device)
+        ext().allgather(_bytes(out), _bytes(xc), _handle(comm))
         return out
 
 
@@ -107,7 +154,7 @@ def alltoall(x, comm):
     with debug_timer("Alltoall", comm.rank, f"{x.numel()} items"):
         xc = x.contiguous()
         out = torch.empty_like(xc)
-        ext().alltoall(_as_real(out), _as_real(xc), _handle(comm))
+        ext().alltoall(_bytes(out), _bytes(xc), _handle(comm))
         return out
 
 
@@ -120,7 +167,7 @@ def bcast(x, root, comm):
     with debug_timer("Bcast", comm.rank, f"{x.numel()} items"):
         xc = x.contiguous()
         out = torch.empty_like(xc)
-        ext().broadcast(_as_real(out), _as_real(xc), root, _handle(comm))
+        ext().broadcast(_bytes(out), _bytes(xc), root, _handle(comm))
         return out
 
 
@@ -128,11 +175,11 @@ def gather(x, root, comm):
     with debug_timer("Gather", comm.rank, f"{x.numel()} items"):
         xc = x.contiguous()
         if comm.rank == root:
-            out = _alloc_out((comm.size,) + tuple(x.shape), x)
+            out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype,
+                              device=x.device)
         else:
-            out = _alloc_out((0,), x)
-        ext().gather(_as_real(out) if out.numel() else out,
-                     _as_real(xc), root, _handle(comm))
+            out = torch.empty(0, dtype=x.dtype, device=x.device)
+        ext().gather(_bytes(out), _bytes(xc), root, _handle(comm))
         return out if comm.rank == root else None
 
 
@@ -140,28 +187,25 @@ def scatter(x, root, comm):
     with debug_timer("Scatter", comm.rank):
         xc = x.contiguous()
         if comm.rank == root:
-            out = _alloc_out(tuple(x.shape[1:]), x)
+            out = torch.empty(tuple(x.shape[1:]), dtype=x.dtype,
+                              device=x.device)
         else:
-            out = _alloc_out(tuple(x.shape), x)
-        ext().scatter(_as_real(out), _as_real(xc), root, _handle(comm))
+            out = torch.empty(tuple(x.shape), dtype=x.dtype, device=x.device)
+        ext().scatter(_bytes(out), _bytes(xc), root, _handle(comm))
         return out
 
 
-def scan(x, op, comm):
-    _check_op(op, x, "scan")
-    if op is Op.AVG:
-        raise ValueError("scan: AVG is not a valid scan operator")
-    with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
-        xc = x.contiguous()
-        out = torch.empty_like(xc)
-        ext().scan(_as_real(out), _as_real(xc), RCCL_OP_ENUM[op],
-                   _handle(comm))
-        return out
+def _self_queue(comm):
+    return _SELF_QUEUES.setdefault(id(comm), [])
 
 
 def send(x, dest, tag, comm):
     with debug_timer("Send", comm.rank, f"to {dest}, tag {tag}"):
-        ext().send(_as_real(x.contiguous()), dest, _handle(comm))
+        xc = x.contiguous()
+        if dest == comm.rank:
+            _self_queue(comm).append((tag, xc.clone()))  # stream-ordered copy
+            return
+        ext().send(_bytes(xc), dest, _handle(comm))
 
 
 def recv(template, source, tag, comm, status):
@@ -172,22 +216,32 @@ def recv(template, source, tag, comm, status):
             "(shapes are static so the source is always known)"
         )
     with debug_timer("Recv", comm.rank, f"from {source}, tag {tag}"):
-        out = torch.empty(
-            tuple(template.shape), dtype=template.dtype,
-            device=template.device,
-        )
-        ext().recv(_as_real(out), source, _handle(comm))
+        out = torch.empty(tuple(template.shape), dtype=template.dtype,
+                          device=template.device)
+        if source == comm.rank:
+            q = _self_queue(comm)
+            for i, (t, buf) in enumerate(q):
+                if tag in (ANY_TAG, t):
+                    q.pop(i)
+                    out.copy_(buf.reshape(out.shape))
+                    _fill_status(status, source, t, out)
+                    return out
+            raise RuntimeError("recv from self with no matching buffered send")
+        ext().recv(_bytes(out), source, _handle(comm))
         _fill_status(status, source, tag, out)
         return out
 
 
 def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
     with debug_timer("Sendrecv", comm.rank, f"src {source} dst {dest}"):
-        out = torch.empty(
-            tuple(recvbuf.shape), dtype=recvbuf.dtype, device=recvbuf.device
-        )
-        ext().sendrecv(_as_real(sendbuf.contiguous()), _as_real(out),
-                       source, dest, _handle(comm))
+        out = torch.empty(tuple(recvbuf.shape), dtype=recvbuf.dtype,
+                          device=recvbuf.device)
+        sc = sendbuf.contiguous()
+        if source == comm.rank and dest == comm.rank:
+            out.copy_(sc.reshape(out.shape))  # stream-ordered device copy
+        else:
+            ext().sendrecv(_bytes(sc), _bytes(out), source, dest,
+                           _handle(comm))
         _fill_status(status, source, recvtag, out)
         return out
 
@@ -197,6 +251,20 @@ def _fill_status(status, source, tag, out):
         status.source = source
         status.tag = tag
         status.count = out.numel() * out.element_size()
+
+
+def pack2d(view):
+    """Gather a 2-D strided device view into a contiguous tensor using the
+    LDS-staged CDNA4 pack kernel (MI355X-native replacement for
+    ``Tensor.contiguous()`` on the halo path)."""
+    out = torch.empty(view.shape, dtype=view.dtype, device=view.device)
+    ext().pack2d(out, view)
+    return out
+
+
+def unpack2d(view, data):
+    """Scatter contiguous ``data`` into a 2-D strided device view."""
+    ext().unpack2d(view, data.contiguous())
 
 
 class group:

@@ -1,0 +1,439 @@
+// mpi4jax_amd native bridge: RCCL collectives enqueued on the current HIP
+// stream, operating zero-copy on torch tensor buffers in HBM3E.
+//
+// This replaces the reference's CUDA FFI backend
+// (/root/reference/mpi4jax/_src/xla_bridge/mpi_xla_bridge_cuda.cpp), whose
+// GPU path was stream-synchronize + host MPI (optionally staging every
+// buffer through host malloc, :185-206).  Here there is NO stream
+// synchronize and NO staging: ncclAllReduce & friends are enqueued directly
+// on the stream torch compute runs on, so ordering against compute is free
+// and the wire rate is xGMI, not PCIe-to-host.
+//
+// Communicator registry: int64 keys -> {ncclComm_t, scratch}, mirroring the
+// reference's int64 handle marshalling idea (mpi_ops_common.h:36-48) but
+// into our own registry instead of raw MPI handles.
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <cstdio>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+
+#include "kernels.h"
+
+namespace {
+
+#define HIP_CHECK(cmd)                                                       \
+  do {                                                                       \
+    hipError_t e_ = (cmd);                                                   \
+    TORCH_CHECK(e_ == hipSuccess, "HIP error: ", hipGetErrorString(e_),      \
+                " at " __FILE__ ":", __LINE__);                              \
+  } while (0)
+
+// Abort-on-error discipline mirrors the reference (mpi_ops_common.h:60-78):
+// a failed collective leaves the communicator unusable, so surface the
+// error loudly.  We throw into Python instead of MPI_Abort so tests can
+// assert on it; an unhandled throw still kills the rank.
+#define RCCL_CHECK(cmd)                                                      \
+  do {                                                                       \
+    ncclResult_t r_ = (cmd);                                                 \
+    TORCH_CHECK(r_ == ncclSuccess, "RCCL error: ", ncclGetErrorString(r_),   \
+                " in " #cmd);                                                \
+  } while (0)
+
+struct CommEntry {
+  ncclComm_t comm = nullptr;
+  int rank = -1;
+  int size = 0;
+  void* barrier_buf = nullptr;  // persistent 4-byte scratch for barrier
+};
+
+std::mutex g_mutex;
+std::unordered_map<int64_t, CommEntry> g_comms;
+int64_t g_next_id = 1;
+bool g_logging = false;
+
+CommEntry& get_comm(int64_t id) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  auto it = g_comms.find(id);
+  TORCH_CHECK(it != g_comms.end(), "unknown RCCL communicator handle ", id);
+  return it->second;
+}
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+ncclDataType_t nccl_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kChar: return ncclInt8;
+    case at::kByte: return ncclUint8;
+    case at::kBool: return ncclUint8;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    case at::kHalf: return ncclFloat16;
+    case at::kFloat: return ncclFloat32;
+    case at::kDouble: return ncclFloat64;
+    case at::kBFloat16: return ncclBfloat16;
+    default:
+      TORCH_CHECK(false, "dtype ", t.scalar_type(),
+                  " not supported by the RCCL backend");
+  }
+}
+
+int dt_code(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return DT_F32;
+    case at::kDouble: return DT_F64;
+    case at::kHalf: return DT_F16;
+    case at::kBFloat16: return DT_BF16;
+    case at::kChar: return DT_I8;
+    case at::kByte: return DT_U8;
+    case at::kBool: return DT_U8;
+    case at::kInt: return DT_I32;
+    case at::kLong: return DT_I64;
+    default:
+      TORCH_CHECK(false, "dtype ", t.scalar_type(),
+                  " not supported by device combine kernels");
+  }
+}
+
+void check_pair(const at::Tensor& out, const at::Tensor& in) {
+  TORCH_CHECK(in.is_cuda() && out.is_cuda(),
+              "RCCL backend requires device tensors");
+  TORCH_CHECK(in.is_contiguous() && out.is_contiguous(),
+              "RCCL backend requires contiguous tensors");
+  TORCH_CHECK(in.scalar_type() == out.scalar_type(), "dtype mismatch");
+}
+
+void log_enqueue(const char* op, const CommEntry& c, int64_t items) {
+  if (g_logging) {
+    std::printf("r%d | native   | %s enqueue (%lld items)\n", c.rank, op,
+                (long long)items);
+    std::fflush(stdout);
+  }
+}
+
+// ---------------------------------------------------------------- lifecycle
+
+py::bytes get_unique_id() {
+  ncclUniqueId id;
+  RCCL_CHECK(ncclGetUniqueId(&id));
+  return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
+}
+
+int64_t comm_init_rank(int64_t nranks, int64_t rank, py::bytes uid_bytes) {
+  std::string uid_str = uid_bytes;
+  TORCH_CHECK(uid_str.size() == sizeof(ncclUniqueId),
+              "bad ncclUniqueId size ", uid_str.size());
+  ncclUniqueId uid;
+  std::memcpy(&uid, uid_str.data(), sizeof(uid));
+  ncclComm_t comm;
+  {
+    // init is collective across ranks; release the GIL so ranks can meet
+    py::gil_scoped_release nogil;
+    RCCL_CHECK(ncclCommInitRank(&comm, (int)nranks, uid, (int)rank));
+  }
+  CommEntry e;
+  e.comm = comm;
+  e.rank = (int)rank;
+  e.size = (int)nranks;
+  HIP_CHECK(hipMalloc(&e.barrier_buf, 8));
+  HIP_CHECK(hipMemset(e.barrier_buf, 0, 8));
+  std::lock_guard<std::mutex> lk(g_mutex);
+  int64_t id = g_next_id++;
+  g_comms[id] = e;
+  return id;
+}
+
+void comm_destroy(int64_t id) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  auto it = g_comms.find(id);
+  if (it == g_comms.end()) return;
+  ncclCommDestroy(it->second.comm);
+  hipFree(it->second.barrier_buf);
+  g_comms.erase(it);
+}
+
+void destroy_all_comms() {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  for (auto& kv : g_comms) {
+    ncclCommDestroy(kv.second.comm);
+    hipFree(kv.second.barrier_buf);
+  }
+  g_comms.clear();
+}
+
+int64_t comm_count() {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  return (int64_t)g_comms.size();
+}
+
+void set_logging(bool enabled) { g_logging = enabled; }
+
+py::dict version_info() {
+  py::dict d;
+  int nccl_ver = 0;
+  ncclGetVersion(&nccl_ver);
+  d["rccl"] = nccl_ver;
+  int hip_ver = 0;
+  hipRuntimeGetVersion(&hip_ver);
+  d["hip_runtime"] = hip_ver;
+  return d;
+}
+
+// --------------------------------------------------------------- collectives
+
+void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  TORCH_CHECK(out.numel() == in.numel(), "size mismatch");
+  log_enqueue("Allreduce", c, in.numel());
+  RCCL_CHECK(ncclAllReduce(in.data_ptr(), out.data_ptr(), in.numel(),
+                           nccl_dtype(in), (ncclRedOp_t)op, c.comm,
+                           cur_stream()));
+}
+
+void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
+            int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  log_enqueue("Reduce", c, in.numel());
+  RCCL_CHECK(ncclReduce(in.data_ptr(), out.data_ptr(), in.numel(),
+                        nccl_dtype(in), (ncclRedOp_t)op, (int)root, c.comm,
+                        cur_stream()));
+}
+
+void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  TORCH_CHECK(out.numel() == in.numel() * c.size, "allgather size mismatch");
+  log_enqueue("Allgather", c, in.numel());
+  RCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), in.numel(),
+                           nccl_dtype(in), c.comm, cur_stream()));
+}
+
+void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  log_enqueue("Bcast", c, in.numel());
+  RCCL_CHECK(ncclBroadcast(in.data_ptr(), out.data_ptr(), in.numel(),
+                           nccl_dtype(in), (int)root, c.comm, cur_stream()));
+}
+
+void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
+                    int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  TORCH_CHECK(in.numel() == out.numel() * c.size,
+              "reduce_scatter size mismatch");
+  log_enqueue("ReduceScatter", c, in.numel());
+  RCCL_CHECK(ncclReduceScatter(in.data_ptr(), out.data_ptr(), out.numel(),
+                               nccl_dtype(in), (ncclRedOp_t)op, c.comm,
+                               cur_stream()));
+}
+
+// grouped p2p composition: RCCL has no alltoall/gather/scatter primitives
+//  — on the fully-connected xGMI clique direct per-peer
+// send/recv IS the bandwidth-optimal algorithm (every peer pair has its own
+// 153 GB/s link; no forwarding needed).
+
+void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  TORCH_CHECK(in.numel() % c.size == 0, "alltoall count not divisible");
+  int64_t chunk = in.numel() / c.size;
+  int64_t esz = in.element_size();
+  auto dt = nccl_dtype(in);
+  char* ip = (char*)in.data_ptr();
+  char* op_ = (char*)out.data_ptr();
+  log_enqueue("Alltoall", c, in.numel());
+  RCCL_CHECK(ncclGroupStart());
+  for (int r = 0; r < c.size; ++r) {
+    RCCL_CHECK(ncclSend(ip + r * chunk * esz, chunk, dt, r, c.comm,
+                        cur_stream()));
+    RCCL_CHECK(ncclRecv(op_ + r * chunk * esz, chunk, dt, r, c.comm,
+                        cur_stream()));
+  }
+  RCCL_CHECK(ncclGroupEnd());
+}
+
+void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad gather input");
+  int64_t chunk = in.numel();
+  int64_t esz = in.element_size();
+  auto dt = nccl_dtype(in);
+  log_enqueue("Gather", c, chunk);
+  RCCL_CHECK(ncclGroupStart());
+  RCCL_CHECK(ncclSend(in.data_ptr(), chunk, dt, (int)root, c.comm,
+                      cur_stream()));
+  if (c.rank == (int)root) {
+    TORCH_CHECK(out.numel() == chunk * c.size, "gather out size mismatch");
+    char* op_ = (char*)out.data_ptr();
+    for (int r = 0; r < c.size; ++r) {
+      RCCL_CHECK(ncclRecv(op_ + r * chunk * esz, chunk, dt, r, c.comm,
+                          cur_stream()));
+    }
+  }
+  RCCL_CHECK(ncclGroupEnd());
+}
+
+void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad scatter output");
+  int64_t chunk = out.numel();
+  int64_t esz = out.element_size();
+  auto dt = nccl_dtype(out);
+  log_enqueue("Scatter", c, chunk);
+  RCCL_CHECK(ncclGroupStart());
+  if (c.rank == (int)root) {
+    TORCH_CHECK(in.numel() == chunk * c.size, "scatter in size mismatch");
+    char* ip = (char*)in.data_ptr();
+    for (int r = 0; r < c.size; ++r) {
+      RCCL_CHECK(ncclSend(ip + r * chunk * esz, chunk, dt, r, c.comm,
+                          cur_stream()));
+    }
+  }
+  RCCL_CHECK(ncclRecv(out.data_ptr(), chunk, dt, (int)root, c.comm,
+                      cur_stream()));
+  RCCL_CHECK(ncclGroupEnd());
+}
+
+void send(at::Tensor in, int64_t dest, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad send input");
+  log_enqueue("Send", c, in.numel());
+  RCCL_CHECK(ncclSend(in.data_ptr(), in.numel(), nccl_dtype(in), (int)dest,
+                      c.comm, cur_stream()));
+}
+
+void recv(at::Tensor out, int64_t source, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad recv output");
+  log_enqueue("Recv", c, out.numel());
+  RCCL_CHECK(ncclRecv(out.data_ptr(), out.numel(), nccl_dtype(out),
+                      (int)source, c.comm, cur_stream()));
+}
+
+void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
+              int64_t dest, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  TORCH_CHECK(sendbuf.is_cuda() && sendbuf.is_contiguous(), "bad sendbuf");
+  TORCH_CHECK(recvbuf.is_cuda() && recvbuf.is_contiguous(), "bad recvbuf");
+  log_enqueue("Sendrecv", c, sendbuf.numel());
+  // grouped => deadlock-free by construction
+  RCCL_CHECK(ncclGroupStart());
+  RCCL_CHECK(ncclSend(sendbuf.data_ptr(), sendbuf.numel(),
+                      nccl_dtype(sendbuf), (int)dest, c.comm, cur_stream()));
+  RCCL_CHECK(ncclRecv(recvbuf.data_ptr(), recvbuf.numel(),
+                      nccl_dtype(recvbuf), (int)source, c.comm,
+                      cur_stream()));
+  RCCL_CHECK(ncclGroupEnd());
+}
+
+void barrier(int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  log_enqueue("Barrier", c, 1);
+  // tiny allreduce on persistent scratch = a cross-rank stream barrier
+  // 
+  RCCL_CHECK(ncclAllReduce(c.barrier_buf, c.barrier_buf, 1, ncclInt32,
+                           ncclSum, c.comm, cur_stream()));
+}
+
+// scan: ring chain with on-device combine.
+// rank r: recv running prefix from r-1, combine with own contribution on
+// the CDNA4 kernel, forward to r+1.  All stream-ordered; the host never
+// blocks.
+void scan(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
+  auto& c = get_comm(comm_id);
+  check_pair(out, in);
+  auto dt = nccl_dtype(in);
+  hipStream_t stream = cur_stream();
+  if (c.rank == 0) {
+    HIP_CHECK(hipMemcpyAsync(out.data_ptr(), in.data_ptr(),
+                             in.numel() * in.element_size(),
+                             hipMemcpyDeviceToDevice, stream));
+  } else {
+    // receive the prefix of ranks [0, r) into out, then out = out (+) in
+    RCCL_CHECK(ncclRecv(out.data_ptr(), out.numel(), dt, c.rank - 1, c.comm,
+                        stream));
+    launch_combine(out.data_ptr(), out.data_ptr(), in.data_ptr(),
+                   in.numel(), dt_code(in), (int)op, stream);
+  }
+  if (c.rank < c.size - 1) {
+    RCCL_CHECK(ncclSend(out.data_ptr(), out.numel(), dt, c.rank + 1, c.comm,
+                        stream));
+  }
+  log_enqueue("Scan", c, in.numel());
+}
+
+void group_start() { RCCL_CHECK(ncclGroupStart()); }
+void group_end() { RCCL_CHECK(ncclGroupEnd()); }
+
+// LDS-staged strided pack/unpack (CDNA4 kernels in kernels.hip).
+// pack2d: gather a 2-D strided view into a contiguous buffer.
+void pack2d(at::Tensor out, at::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && out.is_cuda(), "device tensors required");
+  TORCH_CHECK(in.dim() == 2, "pack2d expects a 2-D view");
+  TORCH_CHECK(out.is_contiguous() && out.numel() == in.numel(),
+              "bad pack2d output");
+  launch_pack2d(out.data_ptr(), in.data_ptr(), in.size(0), in.size(1),
+                in.stride(0), in.stride(1), (int)in.element_size(),
+                cur_stream());
+}
+
+// unpack2d: scatter a contiguous buffer into a 2-D strided view.
+void unpack2d(at::Tensor out, at::Tensor in) {
+  TORCH_CHECK(in.is_cuda() && out.is_cuda(), "device tensors required");
+  TORCH_CHECK(out.dim() == 2, "unpack2d expects a 2-D view");
+  TORCH_CHECK(in.is_contiguous() && out.numel() == in.numel(),
+              "bad unpack2d input");
+  launch_unpack2d(out.data_ptr(), in.data_ptr(), out.size(0), out.size(1),
+                  out.stride(0), out.stride(1), (int)out.element_size(),
+                  cur_stream());
+}
+
+// direct access to the combine kernel (used by gpu numerics tests)
+void combine(at::Tensor dst, at::Tensor a, at::Tensor b, int64_t op) {
+  check_pair(dst, a);
+  check_pair(dst, b);
+  launch_combine(dst.data_ptr(), a.data_ptr(), b.data_ptr(), a.numel(),
+                 dt_code(a), (int)op, cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "mpi4jax_amd native RCCL/HIP bridge (gfx950)";
+  m.def("get_unique_id", &get_unique_id);
+  m.def("comm_init_rank", &comm_init_rank);
+  m.def("comm_destroy", &comm_destroy);
+  m.def("destroy_all_comms", &destroy_all_comms);
+  m.def("comm_count", &comm_count);
+  m.def("set_logging", &set_logging);
+  m.def("version_info", &version_info);
+  m.def("allreduce", &allreduce);
+  m.def("reduce", &reduce);
+  m.def("allgather", &allgather);
+  m.def("broadcast", &broadcast);
+  m.def("reduce_scatter", &reduce_scatter);
+  m.def("alltoall", &alltoall);
+  m.def("gather", &gather);
+  m.def("scatter", &scatter);
+  m.def("send", &send);
+  m.def("recv", &recv);
+  m.def("sendrecv", &sendrecv);
+  m.def("barrier", &barrier);
+  m.def("scan", &scan);
+  m.def("group_start", &group_start);
+  m.def("group_end", &group_end);
+  m.def("pack2d", &pack2d);
+  m.def("unpack2d", &unpack2d);
+  m.def("combine", &combine);
+}

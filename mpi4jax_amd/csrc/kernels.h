@@ -1,0 +1,38 @@
+// CDNA4 device kernels (gfx950) — C ABI, torch-free.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+// dtype codes shared between bridge.cpp and kernels.hip
+enum DtCode : int {
+  DT_F32 = 0,
+  DT_F64 = 1,
+  DT_F16 = 2,
+  DT_BF16 = 3,
+  DT_I8 = 4,
+  DT_U8 = 5,
+  DT_I32 = 6,
+  DT_I64 = 7,
+};
+
+enum OpCode : int {  // matches ncclRedOp_t for the first four
+  OPC_SUM = 0,
+  OPC_PROD = 1,
+  OPC_MAX = 2,
+  OPC_MIN = 3,
+};
+
+// dst[i] = a[i] (op) b[i], grid-strided, enqueued on `stream`
+void launch_combine(void* dst, const void* a, const void* b, long long n,
+                    int dt_code, int op_code, hipStream_t stream);
+
+// out[r*cols + c] = in[r*stride0 + c*stride1]   (strides in ELEMENTS)
+// LDS-staged tiles when neither axis is contiguous on the write side.
+void launch_pack2d(void* out, const void* in, long long rows, long long cols,
+                   long long stride0, long long stride1, int elem_size,
+                   hipStream_t stream);
+
+// in is contiguous (rows, cols); out[r*stride0 + c*stride1] = in[r*cols + c]
+void launch_unpack2d(void* out, const void* in, long long rows,
+                     long long cols, long long stride0, long long stride1,
+                     int elem_size, hipStream_t stream);
