@@ -399,6 +399,46 @@ void unpack2d(at::Tensor out, at::Tensor in) {
                   cur_stream());
 }
 
+// fused shallow-water stage launcher.  bufs order:
+// fe, fn, q, ke, h, u, v, dnh, dnu, dnv, doh, dou, dov
+void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
+              double dy, double dt, double nu, double cor_base,
+              double cor_dj, double ab_a, double ab_b,
+              std::vector<int64_t> flags) {
+  TORCH_CHECK(bufs.size() == 13, "sw_stage expects 13 buffers");
+  TORCH_CHECK(flags.size() == 6, "sw_stage expects 6 flags");
+  const at::Tensor& h = bufs[4];
+  TORCH_CHECK(h.is_cuda() && h.is_contiguous() && h.dim() == 2,
+              "bad shallow-water state tensor");
+  bool is_double = h.scalar_type() == at::kDouble;
+  TORCH_CHECK(is_double || h.scalar_type() == at::kFloat,
+              "shallow-water kernels support f32/f64");
+  SwLaunchParams p;
+  void** slots[13] = {&p.fe, &p.fn, &p.q, &p.ke, &p.h, &p.u, &p.v,
+                      &p.dnh, &p.dnu, &p.dnv, &p.doh, &p.dou, &p.dov};
+  for (int k = 0; k < 13; ++k) {
+    *slots[k] = bufs[k].defined() && bufs[k].numel() ? bufs[k].data_ptr()
+                                                     : nullptr;
+  }
+  p.ny = h.size(0);
+  p.nx = h.size(1);
+  p.dx = dx;
+  p.dy = dy;
+  p.dt = dt;
+  p.nu = nu;
+  p.cor_base = cor_base;
+  p.cor_dj = cor_dj;
+  p.ab_a = ab_a;
+  p.ab_b = ab_b;
+  p.south_open = (int)flags[0];
+  p.north_open = (int)flags[1];
+  p.west_open = (int)flags[2];
+  p.east_open = (int)flags[3];
+  p.east_wall = (int)flags[4];
+  p.north_wall = (int)flags[5];
+  launch_sw_stage((int)stage, p, is_double ? 1 : 0, cur_stream());
+}
+
 // direct access to the combine kernel (used by gpu numerics tests)
 void combine(at::Tensor dst, at::Tensor a, at::Tensor b, int64_t op) {
   check_pair(dst, a);
@@ -436,4 +476,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack2d", &pack2d);
   m.def("unpack2d", &unpack2d);
   m.def("combine", &combine);
+  m.def("sw_stage", &sw_stage);
 }

@@ -36,3 +36,20 @@ void launch_pack2d(void* out, const void* in, long long rows, long long cols,
 void launch_unpack2d(void* out, const void* in, long long rows,
                      long long cols, long long stride0, long long stride1,
                      int elem_size, hipStream_t stream);
+
+// fused shallow-water step stages (shallow_water.hip)
+struct SwLaunchParams {
+  void *fe, *fn, *q, *ke;
+  void *h, *u, *v;
+  void *dnh, *dnu, *dnv;
+  void *doh, *dou, *dov;
+  long long ny, nx;
+  double dx, dy, dt, nu;
+  double cor_base, cor_dj;
+  double ab_a, ab_b;
+  int south_open, north_open, west_open, east_open;
+  int east_wall, north_wall;
+};
+
+void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
+                     hipStream_t stream);

@@ -110,16 +110,28 @@ class CartesianGrid:
         Returns a NEW tensor (the input is never mutated — the reference's
         immutability contract).  Closed-edge halos are left as they were.
         """
+        return self._halo_exchange_impl(arr.clone())
+
+    def halo_exchange_(self, arr):
+        """In-place halo exchange (internal fast path for owned buffers)."""
+        return self._halo_exchange_impl(arr)
+
+    def _halo_exchange_impl(self, out):
         # imported here to avoid a circular import at package init
         from ..ops.sendrecv import sendrecv
         from ..ops.send import send
         from ..ops.recv import recv
 
-        out = arr.clone()
+        me = self.comm.rank
         for send_dir, recv_dir in self._ORDER:
             send_to = self.neighbor(send_dir)
             recv_from = self.neighbor(recv_dir)
             if send_to is None and recv_from is None:
+                continue
+            if send_to == me and recv_from == me:
+                # periodic self-wrap: a plain on-device copy
+                self._set_edge(out, recv_dir,
+                               self._get_edge(out, send_dir, "send"))
                 continue
             if send_to is None:
                 got = recv(self._get_edge(out, recv_dir, "recv"),

@@ -65,7 +65,7 @@ def build_native(force=False):
                                                f"-I{CSRC}"]
 
     objs = []
-    for src in ["kernels.hip", "bridge.cpp"]:
+    for src in ["kernels.hip", "shallow_water.hip", "bridge.cpp"]:
         sp = CSRC / src
         op = BUILD / (src.replace(".", "_") + ".o")
         objs.append(op)
