@@ -405,7 +405,7 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
               double dy, double dt, double nu, double cor_base,
               double cor_dj, double ab_a, double ab_b,
               std::vector<int64_t> flags) {
-  TORCH_CHECK(bufs.size() == 13, "sw_stage expects 13 buffers");
+  TORCH_CHECK(bufs.size() == 16, "sw_stage expects 16 buffers");
   TORCH_CHECK(flags.size() == 6, "sw_stage expects 6 flags");
   const at::Tensor& h = bufs[4];
   TORCH_CHECK(h.is_cuda() && h.is_contiguous() && h.dim() == 2,
@@ -414,9 +414,10 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
   TORCH_CHECK(is_double || h.scalar_type() == at::kFloat,
               "shallow-water kernels support f32/f64");
   SwLaunchParams p;
-  void** slots[13] = {&p.fe, &p.fn, &p.q, &p.ke, &p.h, &p.u, &p.v,
-                      &p.dnh, &p.dnu, &p.dnv, &p.doh, &p.dou, &p.dov};
-  for (int k = 0; k < 13; ++k) {
+  void** slots[16] = {&p.fe, &p.fn, &p.q, &p.ke, &p.h, &p.u, &p.v,
+                      &p.dnh, &p.dnu, &p.dnv, &p.doh, &p.dou, &p.dov,
+                      &p.h2, &p.u2, &p.v2};
+  for (int k = 0; k < 16; ++k) {
     *slots[k] = bufs[k].defined() && bufs[k].numel() ? bufs[k].data_ptr()
                                                      : nullptr;
   }

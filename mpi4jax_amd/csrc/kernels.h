@@ -43,6 +43,7 @@ struct SwLaunchParams {
   void *h, *u, *v;
   void *dnh, *dnu, *dnv;
   void *doh, *dou, *dov;
+  void *h2, *u2, *v2;  // double-buffer outputs for the merged stages
   long long ny, nx;
   double dx, dy, dt, nu;
   double cor_base, cor_dj;

@@ -47,6 +47,9 @@ struct SwArgs {
   T* doh;
   T* dou;
   T* dov;
+  T* h2;
+  T* u2;
+  T* v2;
   long long ny, nx;
   T dx, dy, dt, nu;
   T cor_base, cor_dj;  // coriolis(j) = cor_base + j * cor_dj
@@ -241,6 +244,83 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
   }
 }
 
+// stage 6 = stage2 + stage3 merged with double-buffered field output:
+// tendencies + AB/Euler update written to h2/u2/v2 — no in-place hazard,
+// one full pass of traffic saved.
+template <typename T>
+__global__ void sw_stage6_kernel(SwArgs<T> a) {
+  long long n = a.ny * a.nx;
+  long long ny = a.ny, nx = a.nx;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < n; idx += (long long)gridDim.x * blockDim.x) {
+    long long j = idx / nx, i = idx % nx;
+    T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+      // halo cells: fields pass through (exchange updates them next),
+      // tendencies stay zero
+      a.h2[idx] = h_;
+      a.u2[idx] = u_;
+      a.v2[idx] = v_;
+      continue;
+    }
+
+    T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
+            (a.fn[idx] - a.fn[idx - nx]) / a.dy;
+    a.dnh[idx] = dnh;
+
+    T dnu = -G * (a.h[idx + 1] - h_) / a.dx +
+            T(0.5) * (a.q[idx] * T(0.5) * (a.fn[idx] + a.fn[idx + 1]) +
+                      a.q[idx - nx] * T(0.5) *
+                          (a.fn[idx - nx] + a.fn[idx - nx + 1]));
+    dnu -= (a.ke[idx + 1] - a.ke[idx]) / a.dx;
+    a.dnu[idx] = dnu;
+
+    T dnv = -G * (a.h[idx + nx] - h_) / a.dy -
+            T(0.5) * (a.q[idx] * T(0.5) * (a.fe[idx] + a.fe[idx + nx]) +
+                      a.q[idx - 1] * T(0.5) *
+                          (a.fe[idx - 1] + a.fe[idx + nx - 1]));
+    dnv -= (a.ke[idx + nx] - a.ke[idx]) / a.dy;
+    a.dnv[idx] = dnv;
+
+    T uu = u_ + a.dt * (a.ab_a * dnu + a.ab_b * a.dou[idx]);
+    T vv = v_ + a.dt * (a.ab_a * dnv + a.ab_b * a.dov[idx]);
+    a.h2[idx] = h_ + a.dt * (a.ab_a * dnh + a.ab_b * a.doh[idx]);
+    if (a.f.east_wall && i == nx - 2) uu = T(0);
+    if (a.f.north_wall && j == ny - 2) vv = T(0);
+    a.u2[idx] = uu;
+    a.v2[idx] = vv;
+  }
+}
+
+// stage 7 = stage4 + stage5 merged: friction Laplacian computed inline
+// (each g-gradient evaluated twice — FP is free, the array pass is not),
+// updated u/v written to u2/v2.
+template <typename T>
+__global__ void sw_stage7_kernel(SwArgs<T> a) {
+  long long n = a.ny * a.nx;
+  long long ny = a.ny, nx = a.nx;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       idx < n; idx += (long long)gridDim.x * blockDim.x) {
+    long long j = idx / nx, i = idx % nx;
+    T u_ = a.u[idx], v_ = a.v[idx];
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+      a.u2[idx] = u_;
+      a.v2[idx] = v_;
+      continue;
+    }
+    T lu = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
+           (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
+    T lv = (gu_of_v(a, j, i) - gu_of_v(a, j, i - 1)) / a.dx +
+           (gv_of_v(a, j, i) - gv_of_v(a, j - 1, i)) / a.dy;
+    T uu = u_ + a.dt * lu;
+    T vv = v_ + a.dt * lv;
+    if (a.f.east_wall && i == nx - 2) uu = T(0);
+    if (a.f.north_wall && j == ny - 2) vv = T(0);
+    a.u2[idx] = uu;
+    a.v2[idx] = vv;
+  }
+}
+
 int sw_grid(long long n) {
   long long blocks = (n + kBlock - 1) / kBlock;
   if (blocks > 4096) blocks = 4096;  // grid-stride beyond this
@@ -267,6 +347,9 @@ static void sw_launch(int stage, const SwLaunchParams& p,
   a.doh = (T*)p.doh;
   a.dou = (T*)p.dou;
   a.dov = (T*)p.dov;
+  a.h2 = (T*)p.h2;
+  a.u2 = (T*)p.u2;
+  a.v2 = (T*)p.v2;
   a.ny = p.ny;
   a.nx = p.nx;
   a.dx = (T)p.dx;
@@ -286,6 +369,8 @@ static void sw_launch(int stage, const SwLaunchParams& p,
     case 3: hipLaunchKernelGGL(sw_stage3_kernel<T>, grid, block, 0, stream, a); break;
     case 4: hipLaunchKernelGGL(sw_stage4_kernel<T>, grid, block, 0, stream, a); break;
     case 5: hipLaunchKernelGGL(sw_stage5_kernel<T>, grid, block, 0, stream, a); break;
+    case 6: hipLaunchKernelGGL(sw_stage6_kernel<T>, grid, block, 0, stream, a); break;
+    case 7: hipLaunchKernelGGL(sw_stage7_kernel<T>, grid, block, 0, stream, a); break;
   }
 }
 

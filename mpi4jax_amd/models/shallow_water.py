@@ -195,14 +195,17 @@ class ShallowWater:
         ]
 
     def _init_fused_buffers(self, state):
-        z = lambda: torch.zeros(
-            (self.ny_local, self.nx_local), dtype=self.dtype,
-            device=self.device,
-        )
+        def z():
+            return torch.zeros(
+                (self.ny_local, self.nx_local), dtype=self.dtype,
+                device=self.device,
+            )
+
         fb = {
             "h": state.h.detach().clone().contiguous(),
             "u": state.u.detach().clone().contiguous(),
             "v": state.v.detach().clone().contiguous(),
+            "h_alt": z(), "u_alt": z(), "v_alt": z(),
             "do_h": state.dh.detach().clone().contiguous(),
             "do_u": state.du.detach().clone().contiguous(),
             "do_v": state.dv.detach().clone().contiguous(),
@@ -210,6 +213,11 @@ class ShallowWater:
             "fe": z(), "fn": z(), "q": z(), "ke": z(),
         }
         self._fb = fb
+
+    def _swap(self, *names):
+        fb = self._fb
+        for k in names:
+            fb[k], fb[f"{k}_alt"] = fb[f"{k}_alt"], fb[k]
 
     def _step_fused(self, state, first_step=False):
         from .._backend import rccl
@@ -224,24 +232,25 @@ class ShallowWater:
         cor_dj = float(self.dy * CORIOLIS_BETA)
         ab_a, ab_b = ((1.0, 0.0) if first_step
                       else (ADAMS_BASHFORTH_A, ADAMS_BASHFORTH_B))
-        bufs = [fb["fe"], fb["fn"], fb["q"], fb["ke"], fb["h"], fb["u"],
-                fb["v"], fb["dn_h"], fb["dn_u"], fb["dn_v"], fb["do_h"],
-                fb["do_u"], fb["do_v"]]
 
         def stage(n):
+            bufs = [fb["fe"], fb["fn"], fb["q"], fb["ke"], fb["h"],
+                    fb["u"], fb["v"], fb["dn_h"], fb["dn_u"], fb["dn_v"],
+                    fb["do_h"], fb["do_u"], fb["do_v"], fb["h_alt"],
+                    fb["u_alt"], fb["v_alt"]]
             ext.sw_stage(n, bufs, self.dx, self.dy, self.dt,
                          self.lateral_viscosity, cor_base, cor_dj, ab_a,
                          ab_b, flags)
 
-        stage(1)
-        stage(2)
-        stage(3)
+        stage(1)          # fe, fn, q, ke (with open-edge halo formulas)
+        stage(6)          # tendencies + time update -> h_alt/u_alt/v_alt
+        self._swap("h", "u", "v")
         self.grid.halo_exchange_(fb["h"])
         self.grid.halo_exchange_(fb["u"])
         self.grid.halo_exchange_(fb["v"])
         if self.lateral_viscosity > 0:
-            stage(4)
-            stage(5)
+            stage(7)      # friction Laplacian update -> u_alt/v_alt
+            self._swap("u", "v")
             self.grid.halo_exchange_(fb["u"])
             self.grid.halo_exchange_(fb["v"])
         # the new tendencies become "old" for the next step
@@ -249,6 +258,63 @@ class ShallowWater:
             fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
         return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"], fb["do_u"],
                           fb["do_v"])
+
+    # ------------------------------------------------------------------
+    def make_stepper(self, state, steps_per_call=2, use_graph=None):
+        """Return ``(advance, current)``: ``advance()`` runs
+        ``steps_per_call`` model steps and returns the current state.
+
+        On the fused GPU path the steps are captured into a single hipGraph
+        (collectives included — RCCL is capture-safe), so a whole multistep
+        replays with one launch from the host.  ``steps_per_call`` must be
+        even (buffer parity).  Falls back to an eager loop if capture is
+        unavailable.
+        """
+        if use_graph is None:
+            use_graph = self.fused and self.device.type == "cuda"
+        if use_graph and steps_per_call % 2:
+            raise ValueError("steps_per_call must be even for graph capture")
+
+        if not use_graph:
+            holder = {"s": state}
+
+            def advance():
+                for _ in range(steps_per_call):
+                    holder["s"] = self.step(holder["s"])
+                return holder["s"]
+
+            return advance, state
+
+        # make sure the fused buffers exist & are warm (2 steps, also
+        # pre-allocating every temp the exchange path uses)
+        state = self.step(state)
+        state = self.step(state)
+        torch.cuda.synchronize()
+        try:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                s = state
+                for _ in range(steps_per_call):
+                    s = self.step(s)
+            final = s
+        except Exception:
+            # capture unsupported (e.g. RCCL version without graph
+            # support) — eager fallback
+            torch.cuda.synchronize()
+            holder = {"s": state}
+
+            def advance():
+                for _ in range(steps_per_call):
+                    holder["s"] = self.step(holder["s"])
+                return holder["s"]
+
+            return advance, state
+
+        def advance():
+            graph.replay()
+            return final
+
+        return advance, state
 
     def _step_eager(self, state, first_step=False):
         h, u, v, dh, du, dv = state
