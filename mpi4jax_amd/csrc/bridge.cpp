@@ -440,6 +440,59 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
   launch_sw_stage((int)stage, p, is_double ? 1 : 0, cur_stream());
 }
 
+// halo helpers ------------------------------------------------------------
+
+void collect_field_ptrs(const std::vector<at::Tensor>& fields,
+                        void* ptrs[3], bool& is_double, int64_t& ny,
+                        int64_t& nx) {
+  TORCH_CHECK(!fields.empty() && fields.size() <= 3,
+              "1..3 halo fields supported");
+  ny = fields[0].size(0);
+  nx = fields[0].size(1);
+  is_double = fields[0].scalar_type() == at::kDouble;
+  for (size_t k = 0; k < fields.size(); ++k) {
+    const auto& f = fields[k];
+    TORCH_CHECK(f.is_cuda() && f.is_contiguous() && f.dim() == 2 &&
+                    f.size(0) == ny && f.size(1) == nx &&
+                    f.scalar_type() == fields[0].scalar_type(),
+                "bad halo field");
+    ptrs[k] = f.data_ptr();
+  }
+}
+
+// side 0: east halo col <- col 1; side 1: west halo col <- col nx-2
+void halo_wrap(std::vector<at::Tensor> fields, int64_t side) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  launch_halo_wrap(ptrs, (int)fields.size(), ny, nx, (int)side,
+                   is_double ? 1 : 0, cur_stream());
+}
+
+void pack_cols(at::Tensor buf, std::vector<at::Tensor> fields,
+               int64_t col) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  TORCH_CHECK(buf.is_cuda() && buf.is_contiguous() &&
+                  buf.numel() >= (int64_t)fields.size() * ny,
+              "bad pack buffer");
+  launch_pack_cols(buf.data_ptr(), ptrs, (int)fields.size(), ny, nx, col,
+                   is_double ? 1 : 0, cur_stream());
+}
+
+void unpack_cols(std::vector<at::Tensor> fields, at::Tensor buf,
+                 int64_t col) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  launch_unpack_cols(ptrs, buf.data_ptr(), (int)fields.size(), ny, nx, col,
+                     is_double ? 1 : 0, cur_stream());
+}
+
 // direct access to the combine kernel (used by gpu numerics tests)
 void combine(at::Tensor dst, at::Tensor a, at::Tensor b, int64_t op) {
   check_pair(dst, a);
@@ -478,4 +531,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("unpack2d", &unpack2d);
   m.def("combine", &combine);
   m.def("sw_stage", &sw_stage);
+  m.def("halo_wrap", &halo_wrap);
+  m.def("pack_cols", &pack_cols);
+  m.def("unpack_cols", &unpack_cols);
 }

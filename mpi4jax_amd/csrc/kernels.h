@@ -54,3 +54,15 @@ struct SwLaunchParams {
 
 void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
                      hipStream_t stream);
+
+// halo-exchange helpers (shallow_water.hip): periodic wrap and multi-field
+// column pack/unpack (up to 3 fields)
+void launch_halo_wrap(void* const* fields, int nf, long long ny,
+                      long long nx, int side, int is_double,
+                      hipStream_t stream);
+void launch_pack_cols(void* buf, void* const* fields, int nf, long long ny,
+                      long long nx, long long col, int is_double,
+                      hipStream_t stream);
+void launch_unpack_cols(void* const* fields, const void* buf, int nf,
+                        long long ny, long long nx, long long col,
+                        int is_double, hipStream_t stream);

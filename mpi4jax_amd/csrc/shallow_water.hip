@@ -10,10 +10,11 @@
 // interior + readable halos with an open/closed mask and the only
 // communication left is the h/u/v ring exchange.
 //
-// All loops are 1-D grid-stride over the row-major (ny, nx) array:
-// consecutive lanes touch consecutive columns -> fully coalesced wave64
-// accesses; blocks are 256 threads (4 waves), grid sized to fill all 8
-// XCDs.
+// Mapping: one cell per thread, blockIdx.x*256 covering columns (so a
+// wave64's lanes touch consecutive columns -> fully coalesced), blockIdx.y
+// covering rows; the benchmark grid launches ~27k workgroups, far above
+// the 256-CU residency needed to fill all 8 XCDs.  Indices are 32-bit
+// (int64 div/mod per element measured 2-4x off the HBM roofline).
 
 #include <hip/hip_runtime.h>
 
@@ -60,22 +61,23 @@ struct SwArgs {
 // hc = h padded by edge replication at closed halos (reference
 // shallow_water.py:278-279 + enforce_boundaries); at open halos hc == h.
 template <typename T>
-__device__ inline T hc_at(const SwArgs<T>& a, long long j, long long i) {
+__device__ inline T hc_at(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
   if (j == 0 && !a.f.south_open) j = 1;
-  if (j == a.ny - 1 && !a.f.north_open) j = a.ny - 2;
+  if (j == ny - 1 && !a.f.north_open) j = ny - 2;
   if (i == 0 && !a.f.west_open) i = 1;
-  if (i == a.nx - 1 && !a.f.east_open) i = a.nx - 2;
-  return a.h[j * a.nx + i];
+  if (i == nx - 1 && !a.f.east_open) i = nx - 2;
+  return a.h[j * nx + i];
 }
 
 // stage 1: mass fluxes fe/fn, potential vorticity q, kinetic energy ke
 template <typename T>
 __global__ void sw_stage1_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     T hcc = hc_at(a, j, i);
 
@@ -123,12 +125,12 @@ __global__ void sw_stage1_kernel(SwArgs<T> a) {
 // stage 2: tendencies dnh/dnu/dnv on the interior
 template <typename T>
 __global__ void sw_stage2_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
             (a.fn[idx] - a.fn[idx - nx]) / a.dy;
@@ -153,12 +155,12 @@ __global__ void sw_stage2_kernel(SwArgs<T> a) {
 // stage 3: Adams-Bashforth (or Euler) time update, own-cell, in place
 template <typename T>
 __global__ void sw_stage3_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     T uu = a.u[idx] + a.dt * (a.ab_a * a.dnu[idx] + a.ab_b * a.dou[idx]);
     T vv = a.v[idx] + a.dt * (a.ab_a * a.dnv[idx] + a.ab_b * a.dov[idx]);
     a.h[idx] += a.dt * (a.ab_a * a.dnh[idx] + a.ab_b * a.doh[idx]);
@@ -172,9 +174,9 @@ __global__ void sw_stage3_kernel(SwArgs<T> a) {
 // stage 4: lateral-friction Laplacians lu/lv (reusing fe/fn as scratch
 // would race; dedicated lu/lv buffers are passed via fe/fn slots)
 template <typename T>
-__device__ inline T gu_of_u(const SwArgs<T>& a, long long j, long long i) {
+__device__ inline T gu_of_u(const SwArgs<T>& a, int j, int i) {
   // gu = nu * du/dx, u-kind halos: west col open-only, others unread
-  long long ny = a.ny, nx = a.nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
   if (i > nx - 2 || (i == 0 && !a.f.west_open) || j < 1 || j > ny - 2)
     return T(0);
   if (a.f.east_wall && i == nx - 2) return T(0);
@@ -182,8 +184,8 @@ __device__ inline T gu_of_u(const SwArgs<T>& a, long long j, long long i) {
 }
 
 template <typename T>
-__device__ inline T gv_of_u(const SwArgs<T>& a, long long j, long long i) {
-  long long ny = a.ny, nx = a.nx;
+__device__ inline T gv_of_u(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
   if (j > ny - 2 || (j == 0 && !a.f.south_open) || i < 1 || i > nx - 2)
     return T(0);
   if (a.f.north_wall && j == ny - 2) return T(0);
@@ -191,8 +193,8 @@ __device__ inline T gv_of_u(const SwArgs<T>& a, long long j, long long i) {
 }
 
 template <typename T>
-__device__ inline T gu_of_v(const SwArgs<T>& a, long long j, long long i) {
-  long long ny = a.ny, nx = a.nx;
+__device__ inline T gu_of_v(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
   if (i > nx - 2 || (i == 0 && !a.f.west_open) || j < 1 || j > ny - 2)
     return T(0);
   if (a.f.east_wall && i == nx - 2) return T(0);
@@ -200,8 +202,8 @@ __device__ inline T gu_of_v(const SwArgs<T>& a, long long j, long long i) {
 }
 
 template <typename T>
-__device__ inline T gv_of_v(const SwArgs<T>& a, long long j, long long i) {
-  long long ny = a.ny, nx = a.nx;
+__device__ inline T gv_of_v(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
   if (j > ny - 2 || (j == 0 && !a.f.south_open) || i < 1 || i > nx - 2)
     return T(0);
   if (a.f.north_wall && j == ny - 2) return T(0);
@@ -210,14 +212,14 @@ __device__ inline T gv_of_v(const SwArgs<T>& a, long long j, long long i) {
 
 template <typename T>
 __global__ void sw_stage4_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
   T* lu = a.fe;  // scratch reuse
   T* lv = a.fn;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     lu[idx] = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
               (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
     lv[idx] = (gu_of_v(a, j, i) - gu_of_v(a, j, i - 1)) / a.dx +
@@ -227,14 +229,14 @@ __global__ void sw_stage4_kernel(SwArgs<T> a) {
 
 template <typename T>
 __global__ void sw_stage5_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
   const T* lu = a.fe;
   const T* lv = a.fn;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     T uu = a.u[idx] + a.dt * lu[idx];
     T vv = a.v[idx] + a.dt * lv[idx];
     if (a.f.east_wall && i == nx - 2) uu = T(0);
@@ -249,11 +251,11 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
 // one full pass of traffic saved.
 template <typename T>
 __global__ void sw_stage6_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
     T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
       // halo cells: fields pass through (exchange updates them next),
@@ -261,7 +263,7 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
       a.h2[idx] = h_;
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      continue;
+      return;
     }
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
@@ -297,16 +299,16 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
 // updated u/v written to u2/v2.
 template <typename T>
 __global__ void sw_stage7_kernel(SwArgs<T> a) {
-  long long n = a.ny * a.nx;
-  long long ny = a.ny, nx = a.nx;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-       idx < n; idx += (long long)gridDim.x * blockDim.x) {
-    long long j = idx / nx, i = idx % nx;
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int j = blockIdx.y;
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      continue;
+      return;
     }
     T lu = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
            (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
@@ -362,7 +364,10 @@ static void sw_launch(int stage, const SwLaunchParams& p,
   a.ab_b = (T)p.ab_b;
   a.f = {p.south_open, p.north_open, p.west_open, p.east_open, p.east_wall,
          p.north_wall};
-  dim3 grid(sw_grid(p.ny * p.nx)), block(kBlock);
+  // one cell per thread: x covers columns (coalesced), y covers rows —
+  // ~27k workgroups at the benchmark size fills all 8 XCDs
+  dim3 grid((unsigned)((p.nx + kBlock - 1) / kBlock), (unsigned)p.ny),
+      block(kBlock);
   switch (stage) {
     case 1: hipLaunchKernelGGL(sw_stage1_kernel<T>, grid, block, 0, stream, a); break;
     case 2: hipLaunchKernelGGL(sw_stage2_kernel<T>, grid, block, 0, stream, a); break;
@@ -380,5 +385,119 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
     sw_launch<double>(stage, p, stream);
   } else {
     sw_launch<float>(stage, p, stream);
+  }
+}
+
+// ---------------------------------------------------------------- halo ops
+// One kernel per exchange phase instead of torch narrow copies (measured
+// ~110 us/step of copyBuffer/elementwise time in the eager halo path).
+
+namespace {
+
+template <typename T>
+struct HaloArgs {
+  T* f0;
+  T* f1;
+  T* f2;
+  int nf, ny, nx;
+  int col;  // for pack/unpack
+};
+
+template <typename T>
+__device__ inline T* halo_field(const HaloArgs<T>& a, int f) {
+  return f == 0 ? a.f0 : (f == 1 ? a.f1 : a.f2);
+}
+
+// periodic self-wrap, side 0: east halo col nx-1 <- col 1 ("send west");
+// side 1: west halo col 0 <- col nx-2 ("send east")
+template <typename T>
+__global__ void halo_wrap_kernel(HaloArgs<T> a, int side) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= a.nf * a.ny) return;
+  int f = t / a.ny, j = t % a.ny;
+  T* p = halo_field(a, f) + (long long)j * a.nx;
+  if (side == 0) {
+    p[a.nx - 1] = p[1];
+  } else {
+    p[0] = p[a.nx - 2];
+  }
+}
+
+template <typename T>
+__global__ void pack_cols_kernel(HaloArgs<T> a, T* buf) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= a.nf * a.ny) return;
+  int f = t / a.ny, j = t % a.ny;
+  buf[t] = halo_field(a, f)[(long long)j * a.nx + a.col];
+}
+
+template <typename T>
+__global__ void unpack_cols_kernel(HaloArgs<T> a, const T* buf) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= a.nf * a.ny) return;
+  int f = t / a.ny, j = t % a.ny;
+  halo_field(a, f)[(long long)j * a.nx + a.col] = buf[t];
+}
+
+template <typename T>
+HaloArgs<T> make_halo_args(void* const* fields, int nf, long long ny,
+                           long long nx, long long col) {
+  HaloArgs<T> a;
+  a.f0 = (T*)fields[0];
+  a.f1 = nf > 1 ? (T*)fields[1] : (T*)fields[0];
+  a.f2 = nf > 2 ? (T*)fields[2] : (T*)fields[0];
+  a.nf = nf;
+  a.ny = (int)ny;
+  a.nx = (int)nx;
+  a.col = (int)col;
+  return a;
+}
+
+int halo_grid(int n) { return (n + kBlock - 1) / kBlock; }
+
+}  // namespace
+
+void launch_halo_wrap(void* const* fields, int nf, long long ny,
+                      long long nx, int side, int is_double,
+                      hipStream_t stream) {
+  int n = nf * (int)ny;
+  if (is_double) {
+    auto a = make_halo_args<double>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(halo_wrap_kernel<double>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, side);
+  } else {
+    auto a = make_halo_args<float>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(halo_wrap_kernel<float>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, side);
+  }
+}
+
+void launch_pack_cols(void* buf, void* const* fields, int nf, long long ny,
+                      long long nx, long long col, int is_double,
+                      hipStream_t stream) {
+  int n = nf * (int)ny;
+  if (is_double) {
+    auto a = make_halo_args<double>(fields, nf, ny, nx, col);
+    hipLaunchKernelGGL(pack_cols_kernel<double>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, (double*)buf);
+  } else {
+    auto a = make_halo_args<float>(fields, nf, ny, nx, col);
+    hipLaunchKernelGGL(pack_cols_kernel<float>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, (float*)buf);
+  }
+}
+
+void launch_unpack_cols(void* const* fields, const void* buf, int nf,
+                        long long ny, long long nx, long long col,
+                        int is_double, hipStream_t stream) {
+  int n = nf * (int)ny;
+  if (is_double) {
+    auto a = make_halo_args<double>(fields, nf, ny, nx, col);
+    hipLaunchKernelGGL(unpack_cols_kernel<double>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, (const double*)buf);
+  } else {
+    auto a = make_halo_args<float>(fields, nf, ny, nx, col);
+    hipLaunchKernelGGL(unpack_cols_kernel<float>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a, (const float*)buf);
   }
 }

@@ -245,19 +245,75 @@ class ShallowWater:
         stage(1)          # fe, fn, q, ke (with open-edge halo formulas)
         stage(6)          # tendencies + time update -> h_alt/u_alt/v_alt
         self._swap("h", "u", "v")
-        self.grid.halo_exchange_(fb["h"])
-        self.grid.halo_exchange_(fb["u"])
-        self.grid.halo_exchange_(fb["v"])
+        self._exchange_fields([fb["h"], fb["u"], fb["v"]])
         if self.lateral_viscosity > 0:
             stage(7)      # friction Laplacian update -> u_alt/v_alt
             self._swap("u", "v")
-            self.grid.halo_exchange_(fb["u"])
-            self.grid.halo_exchange_(fb["v"])
+            self._exchange_fields([fb["u"], fb["v"]])
         # the new tendencies become "old" for the next step
         for k in ("h", "u", "v"):
             fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
         return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"], fb["do_u"],
                           fb["do_v"])
+
+    # ------------------------------------------------------------------
+    # fused halo exchange: one CDNA4 kernel per phase (wrap / column
+    # pack+unpack) + grouped RCCL p2p; row phases send/recv directly from
+    # field memory (rows are contiguous — true zero-copy).  Phase order
+    # matches the eager path (W,N,E,S) so halo corners compose identically.
+    def _exchange_fields(self, fields):
+        from .._backend import rccl
+
+        ext = rccl.ext()
+        g = self.grid
+        me = self.comm.rank
+        nf = len(fields)
+        fb = self._fb
+        if "col_sbuf" not in fb or fb["col_sbuf"].numel() < 3 * self.ny_local:
+            fb["col_sbuf"] = torch.empty(3 * self.ny_local, dtype=self.dtype,
+                                         device=self.device)
+            fb["col_rbuf"] = torch.empty_like(fb["col_sbuf"])
+        nx = self.nx_local
+        phases = (
+            ("west", "east", nx - 1, 1),   # recv into east halo, send col 1
+            ("north", "south", 0, -2),     # rows: recv row 0, send row -2
+            ("east", "west", 0, nx - 2),   # recv into west halo, send nx-2
+            ("south", "north", -1, 1),     # rows: recv row -1, send row 1
+        )
+        for k, (send_dir, recv_dir, recv_idx, send_idx) in enumerate(phases):
+            send_to = g.neighbor(send_dir)
+            recv_from = g.neighbor(recv_dir)
+            if send_to is None and recv_from is None:
+                continue
+            cols_phase = k % 2 == 0
+            if cols_phase:
+                if send_to == me and recv_from == me:
+                    # periodic self-wrap: one kernel for all fields
+                    ext.halo_wrap(fields, 0 if recv_idx == nx - 1 else 1)
+                    continue
+                sbuf = fb["col_sbuf"][:nf * self.ny_local]
+                rbuf = fb["col_rbuf"][:nf * self.ny_local]
+                if send_to is not None:
+                    ext.pack_cols(sbuf, fields, send_idx)
+                comm_id = self.comm.rccl_handle()
+                ext.group_start()
+                if send_to is not None:
+                    ext.send(sbuf, send_to, comm_id)
+                if recv_from is not None:
+                    ext.recv(rbuf, recv_from, comm_id)
+                ext.group_end()
+                if recv_from is not None:
+                    ext.unpack_cols(fields, rbuf, recv_idx)
+            else:
+                # rows are contiguous: RCCL moves them in place, zero-copy
+                comm_id = self.comm.rccl_handle()
+                ext.group_start()
+                for f in fields:
+                    if send_to is not None:
+                        ext.send(f[send_idx], send_to, comm_id)
+                    if recv_from is not None:
+                        ext.recv(f[recv_idx], recv_from, comm_id)
+                ext.group_end()
 
     # ------------------------------------------------------------------
     def make_stepper(self, state, steps_per_call=2, use_graph=None):
