@@ -23,6 +23,8 @@
 namespace {
 
 constexpr int kBlock = 256;
+// cells per thread: unrolled constant-offset accesses merge into dwordx4
+constexpr int SW_IPT = 4;
 constexpr float G = 9.81f;
 
 struct SwFlags {
@@ -74,9 +76,13 @@ __device__ inline T hc_at(const SwArgs<T>& a, int j, int i) {
 template <typename T>
 __global__ void sw_stage1_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     T hcc = hc_at(a, j, i);
@@ -126,11 +132,15 @@ __global__ void sw_stage1_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage2_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
             (a.fn[idx] - a.fn[idx - nx]) / a.dy;
@@ -156,11 +166,15 @@ __global__ void sw_stage2_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage3_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
     T uu = a.u[idx] + a.dt * (a.ab_a * a.dnu[idx] + a.ab_b * a.dou[idx]);
     T vv = a.v[idx] + a.dt * (a.ab_a * a.dnv[idx] + a.ab_b * a.dov[idx]);
     a.h[idx] += a.dt * (a.ab_a * a.dnh[idx] + a.ab_b * a.doh[idx]);
@@ -215,11 +229,15 @@ __global__ void sw_stage4_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   T* lu = a.fe;  // scratch reuse
   T* lv = a.fn;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
     lu[idx] = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
               (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
     lv[idx] = (gu_of_v(a, j, i) - gu_of_v(a, j, i - 1)) / a.dx +
@@ -232,11 +250,15 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const T* lu = a.fe;
   const T* lv = a.fn;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
     T uu = a.u[idx] + a.dt * lu[idx];
     T vv = a.v[idx] + a.dt * lv[idx];
     if (a.f.east_wall && i == nx - 2) uu = T(0);
@@ -252,9 +274,13 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage6_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
     T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
@@ -263,7 +289,7 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
       a.h2[idx] = h_;
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      return;
+      continue;
     }
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
@@ -300,15 +326,19 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage7_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int i = blockIdx.x * blockDim.x + threadIdx.x;
   const int j = blockIdx.y;
-  if (i < nx && j < ny) {
+  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
+  if (j >= ny) return;
+#pragma unroll
+  for (int c = 0; c < SW_IPT; ++c) {
+    const int i = i0 + c;
+    if (i >= nx) break;
     const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      return;
+      continue;
     }
     T lu = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
            (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
@@ -366,7 +396,8 @@ static void sw_launch(int stage, const SwLaunchParams& p,
          p.north_wall};
   // one cell per thread: x covers columns (coalesced), y covers rows —
   // ~27k workgroups at the benchmark size fills all 8 XCDs
-  dim3 grid((unsigned)((p.nx + kBlock - 1) / kBlock), (unsigned)p.ny),
+  const long long span = (long long)kBlock * SW_IPT;
+  dim3 grid((unsigned)((p.nx + span - 1) / span), (unsigned)p.ny),
       block(kBlock);
   switch (stage) {
     case 1: hipLaunchKernelGGL(sw_stage1_kernel<T>, grid, block, 0, stream, a); break;
