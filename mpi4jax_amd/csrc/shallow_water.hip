@@ -23,8 +23,23 @@
 namespace {
 
 constexpr int kBlock = 256;
-// cells per thread: unrolled constant-offset accesses merge into dwordx4
-constexpr int SW_IPT = 4;
+
+// XCD-aware block mapping: the dispatcher places block b on XCD b%8, so a
+// plain (col-chunk, row) 2-D grid scatters adjacent rows across XCDs and
+// every j±1 stencil read misses that XCD's L2.  The bijective remap below
+// gives each XCD a contiguous band of row-major tiles, so neighbor rows
+// are L2-resident (MI355X_MICROARCH.md §Workgroup dispatch; speed-only).
+#define SW_BLOCK_MAP(ny, nx)                                               \
+  const int gx_ = (nx + kBlock - 1) / kBlock;                              \
+  const int T_ = (int)gridDim.x;                                           \
+  const int t_ = (int)blockIdx.x;                                          \
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = t_ % 8, yc_ = t_ / 8;        \
+  const int id_ =                                                          \
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)  \
+      + yc_;                                                               \
+  const int j = id_ / gx_;                                                 \
+  const int i = (id_ % gx_) * kBlock + (int)threadIdx.x
+
 constexpr float G = 9.81f;
 
 struct SwFlags {
@@ -76,13 +91,8 @@ __device__ inline T hc_at(const SwArgs<T>& a, int j, int i) {
 template <typename T>
 __global__ void sw_stage1_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     T hcc = hc_at(a, j, i);
@@ -132,15 +142,10 @@ __global__ void sw_stage1_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage2_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
             (a.fn[idx] - a.fn[idx - nx]) / a.dy;
@@ -166,15 +171,10 @@ __global__ void sw_stage2_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage3_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     T uu = a.u[idx] + a.dt * (a.ab_a * a.dnu[idx] + a.ab_b * a.dou[idx]);
     T vv = a.v[idx] + a.dt * (a.ab_a * a.dnv[idx] + a.ab_b * a.dov[idx]);
     a.h[idx] += a.dt * (a.ab_a * a.dnh[idx] + a.ab_b * a.doh[idx]);
@@ -229,15 +229,10 @@ __global__ void sw_stage4_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   T* lu = a.fe;  // scratch reuse
   T* lv = a.fn;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     lu[idx] = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
               (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
     lv[idx] = (gu_of_v(a, j, i) - gu_of_v(a, j, i - 1)) / a.dx +
@@ -250,15 +245,10 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const T* lu = a.fe;
   const T* lv = a.fn;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) continue;
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
     T uu = a.u[idx] + a.dt * lu[idx];
     T vv = a.v[idx] + a.dt * lv[idx];
     if (a.f.east_wall && i == nx - 2) uu = T(0);
@@ -274,13 +264,8 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage6_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
     T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
@@ -289,7 +274,7 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
       a.h2[idx] = h_;
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      continue;
+      return;
     }
 
     T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
@@ -326,19 +311,14 @@ __global__ void sw_stage6_kernel(SwArgs<T> a) {
 template <typename T>
 __global__ void sw_stage7_kernel(SwArgs<T> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int j = blockIdx.y;
-  const int i0 = (blockIdx.x * blockDim.x + threadIdx.x) * SW_IPT;
-  if (j >= ny) return;
-#pragma unroll
-  for (int c = 0; c < SW_IPT; ++c) {
-    const int i = i0 + c;
-    if (i >= nx) break;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
     const int idx = j * nx + i;
     T u_ = a.u[idx], v_ = a.v[idx];
     if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
       a.u2[idx] = u_;
       a.v2[idx] = v_;
-      continue;
+      return;
     }
     T lu = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
            (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
@@ -394,11 +374,10 @@ static void sw_launch(int stage, const SwLaunchParams& p,
   a.ab_b = (T)p.ab_b;
   a.f = {p.south_open, p.north_open, p.west_open, p.east_open, p.east_wall,
          p.north_wall};
-  // one cell per thread: x covers columns (coalesced), y covers rows —
-  // ~27k workgroups at the benchmark size fills all 8 XCDs
-  const long long span = (long long)kBlock * SW_IPT;
-  dim3 grid((unsigned)((p.nx + span - 1) / span), (unsigned)p.ny),
-      block(kBlock);
+  // one cell per thread; 1-D grid of row-major column-chunk tiles with
+  // the XCD-aware remap applied inside the kernel (SW_BLOCK_MAP)
+  long long gx = (p.nx + kBlock - 1) / kBlock;
+  dim3 grid((unsigned)(gx * p.ny)), block(kBlock);
   switch (stage) {
     case 1: hipLaunchKernelGGL(sw_stage1_kernel<T>, grid, block, 0, stream, a); break;
     case 2: hipLaunchKernelGGL(sw_stage2_kernel<T>, grid, block, 0, stream, a); break;
