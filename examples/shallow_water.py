@@ -1,0 +1,98 @@
+"""mpi4jax_amd demo application — distributed shallow-water solver.
+
+The MI355X-native counterpart of the reference demo
+(``/root/reference/examples/shallow_water.py``): a nonlinear shallow-water
+model decomposed over a 2-D process grid, exchanging 1-cell halos through
+mpi4jax_amd ``sendrecv``/``send``/``recv`` every step (fused CDNA4 kernel
+path + hipGraph multistep on GPU).
+
+Usage:
+
+    # single process (CPU or one GPU)
+    python examples/shallow_water.py
+
+    # 4 processes, one per GPU
+    python -m mpi4jax_amd.run -n 4 examples/shallow_water.py
+
+    # benchmark mode: the reference's benchmark config (3600x1800, 0.1
+    # model day, docs/shallow-water.rst:49-52)
+    python -m mpi4jax_amd.run -n 4 examples/shallow_water.py --benchmark
+"""
+
+import argparse
+import sys
+import time
+
+import torch
+
+import mpi4jax_amd as m
+from mpi4jax_amd.models import ShallowWater
+from mpi4jax_amd.models.shallow_water import DAY_IN_SECONDS
+from mpi4jax_amd.parallel.grid import default_dims
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--benchmark", action="store_true",
+                   help="large domain, 0.1 model day, no output")
+    p.add_argument("--days", type=float, default=None,
+                   help="model days to simulate")
+    p.add_argument("--nx", type=int, default=None)
+    p.add_argument("--ny", type=int, default=None)
+    p.add_argument("--save-animation", action="store_true")
+    args = p.parse_args()
+
+    m.init()
+    comm = m.get_world()
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    if args.benchmark:
+        nx, ny = args.nx or 3600, args.ny or 1800
+        days = args.days or 0.1
+    else:
+        nx, ny = args.nx or 360, args.ny or 180
+        days = args.days or 10.0
+
+    sw = ShallowWater(nx=nx, ny=ny, comm=comm, dims=default_dims(comm.size),
+                      device=device)
+    if comm.rank == 0:
+        print(f"# {ny}x{nx} domain on {comm.size} proc(s) "
+              f"({sw.grid.nproc_y}x{sw.grid.nproc_x}), device={device}, "
+              f"dt={sw.dt:.2f}s, {days} model day(s)")
+
+    t0 = time.perf_counter()
+    if args.save_animation:
+        state, steps, wall, _sol = sw.solve(
+            days * DAY_IN_SECONDS, num_multisteps=100, collect=True
+        )
+    else:
+        state, steps, wall = sw.solve(
+            days * DAY_IN_SECONDS, num_multisteps=100
+        )
+    total = time.perf_counter() - t0
+
+    if comm.rank == 0:
+        per_day = wall / days
+        print(f"Solution took {wall:.2f}s "
+              f"({per_day:.2f} s/model-day, {steps / wall:.1f} steps/s, "
+              f"total incl. init {total:.2f}s)")
+
+    if args.save_animation and comm.rank == 0:
+        try:
+            import matplotlib  # noqa
+        except ImportError:
+            print("matplotlib not available; skipping animation",
+                  file=sys.stderr)
+            return
+        # gather + animate intentionally minimal: plot final height field
+        h = sw.gather_global(state.h)
+        import matplotlib.pyplot as plt
+
+        plt.imshow(h.cpu() - 100.0, cmap="RdBu_r", vmin=-10, vmax=10)
+        plt.colorbar(label="surface height anomaly (m)")
+        plt.savefig("shallow-water.png", dpi=100)
+        print("wrote shallow-water.png")
+
+
+if __name__ == "__main__":
+    main()
