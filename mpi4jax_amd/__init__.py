@@ -58,6 +58,7 @@ from .ops import (  # noqa: F401
     gather,
     recv,
     reduce,
+    reduce_scatter,
     scan,
     scatter,
     send,
@@ -73,6 +74,7 @@ __all__ = [
     "gather",
     "recv",
     "reduce",
+    "reduce_scatter",
     "scan",
     "scatter",
     "send",

@@ -35,6 +35,24 @@ CORIOLIS_BETA = 2e-11
 ADAMS_BASHFORTH_A = 1.5 + 0.1
 ADAMS_BASHFORTH_B = -(0.5 + 0.1)
 
+def halo_phase_schedule(nx, ny):
+    """The fused halo exchange's phase plan.
+
+    Mirrors the eager W,N,E,S pairing of :meth:`CartesianGrid.halo_exchange`
+    (and the reference demo, shallow_water.py:180-194) so halo corners
+    compose identically.  Each entry:
+    ``(is_cols_phase, send_dir, recv_dir, recv_index, send_index)`` where
+    indices are columns for cols phases and rows otherwise.
+    Validated against the eager exchange in tests/test_shallow_water.py.
+    """
+    return (
+        (True, "west", "east", nx - 1, 1),   # recv east halo, send col 1
+        (False, "north", "south", 0, ny - 2),  # recv row 0, send row ny-2
+        (True, "east", "west", 0, nx - 2),   # recv west halo, send nx-2
+        (False, "south", "north", ny - 1, 1),  # recv row ny-1, send row 1
+    )
+
+
 _I = slice(1, -1)  # interior
 _L = slice(None, -2)  # shifted left/down
 _R = slice(2, None)  # shifted right/up
@@ -273,19 +291,12 @@ class ShallowWater:
             fb["col_sbuf"] = torch.empty(3 * self.ny_local, dtype=self.dtype,
                                          device=self.device)
             fb["col_rbuf"] = torch.empty_like(fb["col_sbuf"])
-        nx = self.nx_local
-        phases = (
-            ("west", "east", nx - 1, 1),   # recv into east halo, send col 1
-            ("north", "south", 0, -2),     # rows: recv row 0, send row -2
-            ("east", "west", 0, nx - 2),   # recv into west halo, send nx-2
-            ("south", "north", -1, 1),     # rows: recv row -1, send row 1
-        )
-        for k, (send_dir, recv_dir, recv_idx, send_idx) in enumerate(phases):
+        for (cols_phase, send_dir, recv_dir, recv_idx,
+             send_idx) in halo_phase_schedule(self.nx_local, self.ny_local):
             send_to = g.neighbor(send_dir)
             recv_from = g.neighbor(recv_dir)
             if send_to is None and recv_from is None:
                 continue
-            cols_phase = k % 2 == 0
             if cols_phase:
                 if send_to == me and recv_from == me:
                     # periodic self-wrap: one kernel for all fields
@@ -327,7 +338,13 @@ class ShallowWater:
         unavailable.
         """
         if use_graph is None:
-            use_graph = self.fused and self.device.type == "cuda"
+            # graphs only at world size 1 for now: RCCL capture works in
+            # principle, but a capture that succeeds yet replays a stale
+            # p2p schedule would corrupt halos silently — keep multi-rank
+            # on the plain fused loop until graph replay is validated on a
+            # multi-GPU box
+            use_graph = (self.fused and self.device.type == "cuda"
+                         and self.comm.size == 1)
         if use_graph and steps_per_call % 2:
             raise ValueError("steps_per_call must be even for graph capture")
 

@@ -7,6 +7,7 @@ from .bcast import bcast  # noqa: F401
 from .gather import gather  # noqa: F401
 from .recv import recv  # noqa: F401
 from .reduce import reduce  # noqa: F401
+from .reduce_scatter import reduce_scatter  # noqa: F401
 from .scan import scan  # noqa: F401
 from .scatter import scatter  # noqa: F401
 from .send import send  # noqa: F401

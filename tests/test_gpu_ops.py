@@ -104,6 +104,12 @@ def test_gather_scatter_reduce(x):
     assert torch.equal(r, x)
 
 
+def test_reduce_scatter_gpu(x):
+    y = m.reduce_scatter(x[None], m.SUM)
+    torch.cuda.synchronize()
+    assert torch.equal(y, x)
+
+
 def test_scan(x):
     y = m.scan(x, m.SUM)
     torch.cuda.synchronize()

@@ -192,3 +192,15 @@ def _world4(rank, ws):
 @pytest.mark.slow
 def test_world4():
     run_multiproc(_world4, 4)
+
+
+def _reduce_scatter(rank, ws):
+    src = torch.stack([torch.full((3,), float(r + 1)) for r in range(ws)])
+    y = m.reduce_scatter(src + rank, m.SUM)
+    # slice `rank` summed over ranks: ws*(rank+1) + sum(r)
+    expect = ws * (rank + 1) + sum(range(ws))
+    assert torch.equal(y, torch.full((3,), float(expect)))
+
+
+def test_reduce_scatter_multiproc():
+    run_multiproc(_reduce_scatter, 2)

@@ -126,3 +126,10 @@ def test_bad_op(x):
 def test_capability_probes():
     assert isinstance(m.has_rccl_support(), bool)
     assert m.has_cuda_support() == m.has_rccl_support()
+
+
+def test_reduce_scatter(x):
+    y = m.reduce_scatter(x[None], m.SUM)
+    assert torch.equal(y, x)
+    with pytest.raises(ValueError):
+        m.reduce_scatter(torch.zeros(3, 2), m.SUM)

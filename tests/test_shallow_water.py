@@ -61,6 +61,78 @@ def test_decomposed_matches_single_2x2():
     run_multiproc(_decomposed_matches_single, 4)
 
 
+def _phase_schedule_worker(rank, ws):
+    """Validate the fused halo phase plan against the eager exchange.
+
+    The fused GPU path (ShallowWater._exchange_fields) follows
+    halo_phase_schedule(); executing the same plan with CPU ops must
+    reproduce CartesianGrid.halo_exchange exactly, corners included.
+    """
+    from mpi4jax_amd.parallel.grid import CartesianGrid
+    from mpi4jax_amd.models.shallow_water import halo_phase_schedule
+
+    for dims, periodic_x in (((2, 2), True), ((2, 2), False),
+                             ((2, 1), True), ((1, 2), True)):
+        if dims[0] * dims[1] != ws:
+            continue
+        g = CartesianGrid(m.get_world(), dims=dims,
+                          periodic=(False, periodic_x))
+        torch.manual_seed(100 + rank)
+        ny, nx = 6, 7
+        f0 = torch.randn(ny, nx)
+        f1 = torch.randn(ny, nx)
+        expect = [g.halo_exchange(f0), g.halo_exchange(f1)]
+
+        fields = [f0.clone(), f1.clone()]
+        me = g.comm.rank
+        for cols, sdir, rdir, ridx, sidx in halo_phase_schedule(nx, ny):
+            st, rf = g.neighbor(sdir), g.neighbor(rdir)
+            if st is None and rf is None:
+                continue
+            if st == me and rf == me:
+                for f in fields:
+                    if cols:
+                        f[:, ridx] = f[:, sidx]
+                    else:
+                        f[ridx, :] = f[sidx, :]
+                continue
+            if cols:
+                sbuf = torch.cat([f[:, sidx] for f in fields])
+                tmpl = torch.empty(len(fields) * ny)
+                if st is not None and rf is not None:
+                    rbuf = m.sendrecv(sbuf, tmpl, source=rf, dest=st,
+                                      comm=g.comm)
+                elif st is not None:
+                    m.send(sbuf, st, comm=g.comm)
+                    rbuf = None
+                else:
+                    rbuf = m.recv(tmpl, rf, comm=g.comm)
+                if rf is not None:
+                    for i, f in enumerate(fields):
+                        f[:, ridx] = rbuf[i * ny:(i + 1) * ny]
+            else:
+                for f in fields:
+                    if st is not None and rf is not None:
+                        got = m.sendrecv(f[sidx, :], f[ridx, :], source=rf,
+                                         dest=st, comm=g.comm)
+                        f[ridx, :] = got
+                    elif st is not None:
+                        m.send(f[sidx, :].contiguous(), st, comm=g.comm)
+                    else:
+                        f[ridx, :] = m.recv(f[ridx, :], rf, comm=g.comm)
+        for got, exp in zip(fields, expect):
+            assert torch.equal(got, exp), (dims, periodic_x, rank,
+                                           (got - exp).abs().max())
+
+
+def test_fused_halo_phase_schedule_2x2():
+    run_multiproc(_phase_schedule_worker, 4)
+
+
+def test_fused_halo_phase_schedule_2ranks():
+    run_multiproc(_phase_schedule_worker, 2)
+
+
 def _two_rank(rank, ws):
     sw = ShallowWater(nx=24, ny=12, comm=m.get_world(), dims=(2, 1))
     state, steps, wall = sw.solve(t1_seconds=sw.dt * 10, num_multisteps=5)
