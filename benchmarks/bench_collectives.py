@@ -21,6 +21,9 @@ import time
 
 import torch
 
+import os as _os, sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import mpi4jax_amd as m
 
 

@@ -25,6 +25,9 @@ import time
 
 import torch
 
+import os as _os, sys as _sys
+_sys.path.insert(0, _os.path.dirname(_os.path.dirname(_os.path.abspath(__file__))))
+
 import mpi4jax_amd as m
 from mpi4jax_amd.models import ShallowWater
 from mpi4jax_amd.models.shallow_water import DAY_IN_SECONDS
