@@ -300,7 +300,9 @@ class ShallowWater:
             if cols_phase:
                 if send_to == me and recv_from == me:
                     # periodic self-wrap: one kernel for all fields
-                    ext.halo_wrap(fields, 0 if recv_idx == nx - 1 else 1)
+                    ext.halo_wrap(
+                        fields, 0 if recv_idx == self.nx_local - 1 else 1
+                    )
                     continue
                 sbuf = fb["col_sbuf"][:nf * self.ny_local]
                 rbuf = fb["col_rbuf"][:nf * self.ny_local]
