@@ -258,6 +258,109 @@ __global__ void sw_stage5_kernel(SwArgs<T> a) {
   }
 }
 
+
+// masked derived-field evaluators: single source of truth for the fe/fn/
+// q/ke formulas AND their open/closed halo masks (same semantics as the
+// stage-1 kernel, which remains as the two-pass fallback).  Evaluating
+// these per use in the merged stage-8 kernel trades cheap cached loads +
+// flops for a full write+read pass of four 26 MB arrays.
+template <typename T>
+__device__ inline T fe_at(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  if (i > nx - 2) return T(0);
+  if (!((j >= 1 || a.f.south_open) && (j <= ny - 2 || a.f.north_open) &&
+        (i >= 1 || a.f.west_open)))
+    return T(0);
+  if (a.f.east_wall && i == nx - 2) return T(0);
+  return T(0.5) * (hc_at(a, j, i) + hc_at(a, j, i + 1)) * a.u[j * nx + i];
+}
+
+template <typename T>
+__device__ inline T fn_at(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  if (j > ny - 2) return T(0);
+  if (!((i >= 1 || a.f.west_open) && (i <= nx - 2 || a.f.east_open) &&
+        (j >= 1 || a.f.south_open)))
+    return T(0);
+  if (a.f.north_wall && j == ny - 2) return T(0);
+  return T(0.5) * (hc_at(a, j, i) + hc_at(a, j + 1, i)) * a.v[j * nx + i];
+}
+
+template <typename T>
+__device__ inline T q_at(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  if (i > nx - 2 || j > ny - 2) return T(0);
+  if (!((j >= 1 || a.f.south_open) && (i >= 1 || a.f.west_open)))
+    return T(0);
+  const int idx = j * nx + i;
+  T cor = a.cor_base + (T)j * a.cor_dj;
+  T q = cor + ((a.v[idx + 1] - a.v[idx]) / a.dx -
+               (a.u[idx + nx] - a.u[idx]) / a.dy);
+  q *= T(1) / (T(0.25) * (hc_at(a, j, i) + hc_at(a, j, i + 1) +
+                          hc_at(a, j + 1, i) + hc_at(a, j + 1, i + 1)));
+  return q;
+}
+
+template <typename T>
+__device__ inline T ke_at(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  if (i < 1 || j < 1) return T(0);
+  if (!((i <= nx - 2 || a.f.east_open) && (j <= ny - 2 || a.f.north_open)))
+    return T(0);
+  const int idx = j * nx + i;
+  T u_ = a.u[idx], um = a.u[idx - 1];
+  T v_ = a.v[idx], vm = a.v[idx - nx];
+  return T(0.5) * (T(0.5) * (u_ * u_ + um * um) +
+                   T(0.5) * (v_ * v_ + vm * vm));
+}
+
+// stage 8 = stage 1 + stage 6 fused: derived fields evaluated in-register,
+// no fe/fn/q/ke array traffic at all.
+template <typename T>
+__global__ void sw_stage8_kernel(SwArgs<T> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  SW_BLOCK_MAP(ny, nx);
+  if (i < nx && j < ny) {
+    const int idx = j * nx + i;
+    T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
+    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+      a.h2[idx] = h_;
+      a.u2[idx] = u_;
+      a.v2[idx] = v_;
+      return;
+    }
+
+    T dnh = -(fe_at(a, j, i) - fe_at(a, j, i - 1)) / a.dx -
+            (fn_at(a, j, i) - fn_at(a, j - 1, i)) / a.dy;
+    a.dnh[idx] = dnh;
+
+    T qc = q_at(a, j, i), qs = q_at(a, j - 1, i), qw = q_at(a, j, i - 1);
+    T fnc = fn_at(a, j, i), fne = fn_at(a, j, i + 1);
+    T fns = fn_at(a, j - 1, i), fnse = fn_at(a, j - 1, i + 1);
+    T dnu = -G * (a.h[idx + 1] - h_) / a.dx +
+            T(0.5) * (qc * T(0.5) * (fnc + fne) +
+                      qs * T(0.5) * (fns + fnse));
+    dnu -= (ke_at(a, j, i + 1) - ke_at(a, j, i)) / a.dx;
+    a.dnu[idx] = dnu;
+
+    T fec = fe_at(a, j, i), fen = fe_at(a, j + 1, i);
+    T few = fe_at(a, j, i - 1), fenw = fe_at(a, j + 1, i - 1);
+    T dnv = -G * (a.h[idx + nx] - h_) / a.dy -
+            T(0.5) * (qc * T(0.5) * (fec + fen) +
+                      qw * T(0.5) * (few + fenw));
+    dnv -= (ke_at(a, j + 1, i) - ke_at(a, j, i)) / a.dy;
+    a.dnv[idx] = dnv;
+
+    T uu = u_ + a.dt * (a.ab_a * dnu + a.ab_b * a.dou[idx]);
+    T vv = v_ + a.dt * (a.ab_a * dnv + a.ab_b * a.dov[idx]);
+    a.h2[idx] = h_ + a.dt * (a.ab_a * dnh + a.ab_b * a.doh[idx]);
+    if (a.f.east_wall && i == nx - 2) uu = T(0);
+    if (a.f.north_wall && j == ny - 2) vv = T(0);
+    a.u2[idx] = uu;
+    a.v2[idx] = vv;
+  }
+}
+
 // stage 6 = stage2 + stage3 merged with double-buffered field output:
 // tendencies + AB/Euler update written to h2/u2/v2 — no in-place hazard,
 // one full pass of traffic saved.
@@ -386,6 +489,7 @@ static void sw_launch(int stage, const SwLaunchParams& p,
     case 5: hipLaunchKernelGGL(sw_stage5_kernel<T>, grid, block, 0, stream, a); break;
     case 6: hipLaunchKernelGGL(sw_stage6_kernel<T>, grid, block, 0, stream, a); break;
     case 7: hipLaunchKernelGGL(sw_stage7_kernel<T>, grid, block, 0, stream, a); break;
+    case 8: hipLaunchKernelGGL(sw_stage8_kernel<T>, grid, block, 0, stream, a); break;
   }
 }
 

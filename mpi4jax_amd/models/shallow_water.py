@@ -260,8 +260,13 @@ class ShallowWater:
                          self.lateral_viscosity, cor_base, cor_dj, ab_a,
                          ab_b, flags)
 
-        stage(1)          # fe, fn, q, ke (with open-edge halo formulas)
-        stage(6)          # tendencies + time update -> h_alt/u_alt/v_alt
+        import os
+
+        if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
+            stage(1)      # fe, fn, q, ke (with open-edge halo formulas)
+            stage(6)      # tendencies + time update -> h_alt/u_alt/v_alt
+        else:
+            stage(8)      # fused: derived fields in-register, no arrays
         self._swap("h", "u", "v")
         self._exchange_fields([fb["h"], fb["u"], fb["v"]])
         if self.lateral_viscosity > 0:
