@@ -262,15 +262,25 @@ class ShallowWater:
 
         import os
 
-        if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
-            stage(1)      # fe, fn, q, ke (with open-edge halo formulas)
-            stage(6)      # tendencies + time update -> h_alt/u_alt/v_alt
+        # float32 runs the vectorized (float4) stage kernels; f64 and the
+        # MPI4JAX_AMD_SW_NOVEC escape hatch run the scalar ones.  Stage 8
+        # (fully merged, derived fields in-register) measured slower than
+        # the two-pass kernels — kept for reference, selectable via
+        # MPI4JAX_AMD_SW_MERGED.
+        if os.environ.get("MPI4JAX_AMD_SW_MERGED"):
+            s1, s6, s7 = None, 8, 7
+        elif (self.dtype == torch.float32
+                and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
+            s1, s6, s7 = 11, 16, 17
         else:
-            stage(8)      # fused: derived fields in-register, no arrays
+            s1, s6, s7 = 1, 6, 7
+        if s1 is not None:
+            stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
+        stage(s6)         # tendencies + time update -> h_alt/u_alt/v_alt
         self._swap("h", "u", "v")
         self._exchange_fields([fb["h"], fb["u"], fb["v"]])
         if self.lateral_viscosity > 0:
-            stage(7)      # friction Laplacian update -> u_alt/v_alt
+            stage(s7)     # friction Laplacian update -> u_alt/v_alt
             self._swap("u", "v")
             self._exchange_fields([fb["u"], fb["v"]])
         # the new tendencies become "old" for the next step

@@ -274,6 +274,32 @@ def test_fused_step_matches_eager_f64_tight():
         )
 
 
+def test_vectorized_stages_match_scalar(monkeypatch):
+    """float4 stage kernels vs scalar stage kernels (same formulas; FP
+    contraction may differ slightly)."""
+    from mpi4jax_amd.models import ShallowWater
+
+    results = {}
+    for novec in ("1", ""):
+        if novec:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_NOVEC", "1")
+        else:
+            monkeypatch.delenv("MPI4JAX_AMD_SW_NOVEC", raising=False)
+        sw = ShallowWater(nx=120, ny=60, device="cuda",
+                          comm=m.get_world().Clone())
+        st = sw.initial_conditions()
+        st = sw.step(st, first_step=True)
+        for _ in range(8):
+            st = sw.step(st)
+        torch.cuda.synchronize()
+        results[novec] = st
+    for name in ("h", "u", "v"):
+        a, b = getattr(results["1"], name), getattr(results[""], name)
+        assert torch.allclose(a, b, atol=1e-4, rtol=1e-5), (
+            name, (a - b).abs().max().item()
+        )
+
+
 def test_graph_stepper_matches_plain_fused():
     """hipGraph-captured multistep must reproduce the plain fused loop."""
     from mpi4jax_amd.models import ShallowWater
