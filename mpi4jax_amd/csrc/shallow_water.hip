@@ -553,10 +553,10 @@ __device__ inline void stage7_cell(const SwArgs<float>& a, int j, int i) {
 __global__ void sw_stage1v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= (long long)ppr * ny) return;
-  const int j = (int)(t / ppr);
-  const int i0 = (int)(t % ppr) * 4;
+  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (t >= ppr * ny) return;
+  const int j = t / ppr;
+  const int i0 = (t % ppr) * 4;
   const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
   const int imax = (a.f.east_open && !a.f.east_wall) ? nx - 1 : nx - 2;
   const int imin = a.f.west_open ? 0 + 1 : 2;  // cols use i-1 only in ke
@@ -585,10 +585,10 @@ __global__ void sw_stage1v(SwArgs<float> a) {
 __global__ void sw_stage6v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= (long long)ppr * ny) return;
-  const int j = (int)(t / ppr);
-  const int i0 = (int)(t % ppr) * 4;
+  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (t >= ppr * ny) return;
+  const int j = t / ppr;
+  const int i0 = (t % ppr) * 4;
   const int jmax = a.f.north_wall ? ny - 3 : ny - 2;
   const int imax = a.f.east_wall ? nx - 2 : nx - 1;
   const bool fast = j >= 1 && j <= jmax && i0 >= 1 && i0 + 4 <= imax;
@@ -628,10 +628,10 @@ __global__ void sw_stage6v(SwArgs<float> a) {
 __global__ void sw_stage7v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const long long t = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (t >= (long long)ppr * ny) return;
-  const int j = (int)(t / ppr);
-  const int i0 = (int)(t % ppr) * 4;
+  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (t >= ppr * ny) return;
+  const int j = t / ppr;
+  const int i0 = (t % ppr) * 4;
   const int jmin = a.f.south_open ? 1 : 2;
   const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
   const int imin = a.f.west_open ? 1 : 2;
