@@ -226,15 +226,17 @@ def test_shallow_water_gpu_step():
     assert torch.allclose(state.h.cpu(), sc.h, atol=1e-4, rtol=1e-4)
 
 
-def test_fused_step_matches_eager():
+@pytest.mark.parametrize("nx,ny", [(120, 60), (37, 19), (50, 26), (41, 23)])
+def test_fused_step_matches_eager(nx, ny):
     """Fused CDNA4 kernel path vs eager torch path (same scheme; fused
-    kernels may contract to FMA, so compare with a small tolerance)."""
+    kernels may contract to FMA, so compare with a small tolerance).
+    Odd sizes exercise the vector kernels' scalar edge fallback."""
     from mpi4jax_amd.models import ShallowWater
 
     n_steps = 10
     results = {}
     for fused in (False, True):
-        sw = ShallowWater(nx=120, ny=60, device="cuda", fused=fused,
+        sw = ShallowWater(nx=nx, ny=ny, device="cuda", fused=fused,
                           comm=m.get_world().Clone())
         st = sw.initial_conditions()
         st = sw.step(st, first_step=True)
