@@ -98,3 +98,23 @@ def test_distributed_transpose_pattern():
     x = torch.arange(12.0).reshape(1, 3, 4)
     y = m.alltoall(x)
     assert torch.equal(y, x)
+
+
+def _distributed_transpose(rank, ws):
+    """Row-partitioned global matrix transposed via alltoall."""
+    p, q = 3, 4 * ws
+    torch.manual_seed(7)
+    M = torch.randn(ws * p, q)
+    local = M[rank * p:(rank + 1) * p]  # (p, q)
+
+    # split columns into per-destination chunks, exchange, reassemble
+    chunks = local.reshape(p, ws, q // ws).permute(1, 0, 2).contiguous()
+    got = m.alltoall(chunks)  # (ws, p, q//ws): block r of rank r's rows
+    mine = got.reshape(ws * p, q // ws).t().contiguous()  # (q//ws, ws*p)
+
+    expect = M.t()[rank * (q // ws):(rank + 1) * (q // ws)]
+    assert torch.equal(mine, expect), (rank, (mine - expect).abs().max())
+
+
+def test_distributed_transpose_multiproc():
+    run_multiproc(_distributed_transpose, 2)
