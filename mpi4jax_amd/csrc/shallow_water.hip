@@ -770,11 +770,8 @@ __global__ void halo_wrap_kernel(HaloArgs<T> a, int side) {
   if (t >= a.nf * a.ny) return;
   int f = t / a.ny, j = t % a.ny;
   T* p = halo_field(a, f) + (long long)j * a.nx;
-  if (side == 0) {
-    p[a.nx - 1] = p[1];
-  } else {
-    p[0] = p[a.nx - 2];
-  }
+  if (side != 1) p[a.nx - 1] = p[1];
+  if (side != 0) p[0] = p[a.nx - 2];  // side 2 = both wraps in one launch
 }
 
 template <typename T>

@@ -38,17 +38,18 @@ ADAMS_BASHFORTH_B = -(0.5 + 0.1)
 def halo_phase_schedule(nx, ny):
     """The fused halo exchange's phase plan.
 
-    Mirrors the eager W,N,E,S pairing of :meth:`CartesianGrid.halo_exchange`
-    (and the reference demo, shallow_water.py:180-194) so halo corners
-    compose identically.  Each entry:
+    Mirrors :meth:`CartesianGrid.halo_exchange`: both column exchanges
+    first (independent of each other), then both row exchanges (full rows,
+    carrying the freshly-filled column-halo corners).  Each entry:
     ``(is_cols_phase, send_dir, recv_dir, recv_index, send_index)`` where
-    indices are columns for cols phases and rows otherwise.
+    indices are columns for cols phases and rows otherwise.  The two
+    phases of each half are batched into one RCCL group by the fused path.
     Validated against the eager exchange in tests/test_shallow_water.py.
     """
     return (
         (True, "west", "east", nx - 1, 1),   # recv east halo, send col 1
-        (False, "north", "south", 0, ny - 2),  # recv row 0, send row ny-2
         (True, "east", "west", 0, nx - 2),   # recv west halo, send nx-2
+        (False, "north", "south", 0, ny - 2),  # recv row 0, send row ny-2
         (False, "south", "north", ny - 1, 1),  # recv row ny-1, send row 1
     )
 
@@ -301,47 +302,65 @@ class ShallowWater:
         g = self.grid
         me = self.comm.rank
         nf = len(fields)
+        ny = self.ny_local
         fb = self._fb
-        if "col_sbuf" not in fb or fb["col_sbuf"].numel() < 3 * self.ny_local:
-            fb["col_sbuf"] = torch.empty(3 * self.ny_local, dtype=self.dtype,
-                                         device=self.device)
-            fb["col_rbuf"] = torch.empty_like(fb["col_sbuf"])
-        for (cols_phase, send_dir, recv_dir, recv_idx,
-             send_idx) in halo_phase_schedule(self.nx_local, self.ny_local):
-            send_to = g.neighbor(send_dir)
-            recv_from = g.neighbor(recv_dir)
-            if send_to is None and recv_from is None:
+        if "col_sbuf0" not in fb or fb["col_sbuf0"].numel() < 3 * ny:
+            for k in ("col_sbuf0", "col_rbuf0", "col_sbuf1", "col_rbuf1"):
+                fb[k] = torch.empty(3 * ny, dtype=self.dtype,
+                                    device=self.device)
+        sched = halo_phase_schedule(self.nx_local, ny)
+
+        # cols half: both column phases in ONE RCCL group (self-wrap runs
+        # as a single kernel and never touches RCCL — keeps the world-1
+        # path graph-capturable with zero comm init)
+        pend = []
+        col_nbrs = [(g.neighbor(sdir), g.neighbor(rdir))
+                    for _, sdir, rdir, _, _ in sched[:2]]
+        if all(st == me and rf == me for st, rf in col_nbrs):
+            ext.halo_wrap(fields, 2)  # both periodic wraps, one kernel
+            col_nbrs = None
+        for k, (_, sdir, rdir, ridx, sidx) in enumerate(sched[:2]):
+            if col_nbrs is None:
+                break
+            st, rf = col_nbrs[k]
+            if st is None and rf is None:
                 continue
-            if cols_phase:
-                if send_to == me and recv_from == me:
-                    # periodic self-wrap: one kernel for all fields
-                    ext.halo_wrap(
-                        fields, 0 if recv_idx == self.nx_local - 1 else 1
-                    )
-                    continue
-                sbuf = fb["col_sbuf"][:nf * self.ny_local]
-                rbuf = fb["col_rbuf"][:nf * self.ny_local]
-                if send_to is not None:
-                    ext.pack_cols(sbuf, fields, send_idx)
-                comm_id = self.comm.rccl_handle()
-                ext.group_start()
-                if send_to is not None:
-                    ext.send(sbuf, send_to, comm_id)
-                if recv_from is not None:
-                    ext.recv(rbuf, recv_from, comm_id)
-                ext.group_end()
-                if recv_from is not None:
-                    ext.unpack_cols(fields, rbuf, recv_idx)
-            else:
-                # rows are contiguous: RCCL moves them in place, zero-copy
-                comm_id = self.comm.rccl_handle()
-                ext.group_start()
+            if st == me and rf == me:
+                ext.halo_wrap(fields,
+                              0 if ridx == self.nx_local - 1 else 1)
+                continue
+            sbuf = fb[f"col_sbuf{k}"][:nf * ny]
+            rbuf = fb[f"col_rbuf{k}"][:nf * ny]
+            if st is not None:
+                ext.pack_cols(sbuf, fields, sidx)
+            pend.append((st, rf, sbuf, rbuf, ridx))
+        if pend:
+            comm_id = self.comm.rccl_handle()
+            ext.group_start()
+            for st, rf, sbuf, rbuf, _ in pend:
+                if st is not None:
+                    ext.send(sbuf, st, comm_id)
+                if rf is not None:
+                    ext.recv(rbuf, rf, comm_id)
+            ext.group_end()
+            for st, rf, _, rbuf, ridx in pend:
+                if rf is not None:
+                    ext.unpack_cols(fields, rbuf, ridx)
+
+        # rows half: full rows are contiguous — RCCL moves them in place
+        # (zero-copy), both directions and all fields in ONE group
+        rows = [(g.neighbor(sdir), g.neighbor(rdir), ridx, sidx)
+                for _, sdir, rdir, ridx, sidx in sched[2:]]
+        if any(st is not None or rf is not None for st, rf, _, _ in rows):
+            comm_id = self.comm.rccl_handle()
+            ext.group_start()
+            for st, rf, ridx, sidx in rows:
                 for f in fields:
-                    if send_to is not None:
-                        ext.send(f[send_idx], send_to, comm_id)
-                    if recv_from is not None:
-                        ext.recv(f[recv_idx], recv_from, comm_id)
-                ext.group_end()
+                    if st is not None:
+                        ext.send(f[sidx], st, comm_id)
+                    if rf is not None:
+                        ext.recv(f[ridx], rf, comm_id)
+            ext.group_end()
 
     # ------------------------------------------------------------------
     def make_stepper(self, state, steps_per_call=2, use_graph=None):

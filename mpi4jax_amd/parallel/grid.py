@@ -80,16 +80,19 @@ class CartesianGrid:
             ix2 %= self.nproc_x
         return self.rank_at(iy2, ix2)
 
-    # halo slice tables for a (ny, nx) array with a 1-cell halo ring;
-    # same send/recv pairing and clockwise order as the reference demo
-    # (shallow_water.py:180-208) so the communication pattern — and its
-    # deadlock-freedom argument — carries over.
+    # halo slice tables for a (ny, nx) array with a 1-cell halo ring.
+    # The reference demo exchanges in clockwise W,N,E,S order
+    # (shallow_water.py:180-208); we use both column exchanges first, then
+    # both row exchanges — full-row sends then carry fresh column-halo
+    # corners, which satisfies every corner dependency the clockwise order
+    # satisfies (and makes the SW corner fresh too), while letting the
+    # fused GPU path batch each half into a single RCCL group.
     _SEND_ROW = {"south": 1, "north": -2}
     _RECV_ROW = {"south": 0, "north": -1}
     _SEND_COL = {"west": 1, "east": -2}
     _RECV_COL = {"west": 0, "east": -1}
-    _ORDER = (("west", "east"), ("north", "south"),
-              ("east", "west"), ("south", "north"))
+    _ORDER = (("west", "east"), ("east", "west"),
+              ("north", "south"), ("south", "north"))
 
     def _get_edge(self, arr, direction, kind):
         if direction in ("south", "north"):
