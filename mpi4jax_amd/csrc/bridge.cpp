@@ -174,6 +174,30 @@ int64_t comm_count() {
   return (int64_t)g_comms.size();
 }
 
+// async-error surfacing: RCCL collectives are enqueued asynchronously, so
+// a transport failure shows up later on the communicator.  The reference's
+// failure model is fail-fast abort (mpi_ops_common.h:60-78); here
+// check_async_errors() is called from flush()/finalize() and can be called
+// by users — it throws if any communicator has failed.
+void check_async_errors() {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  for (auto& kv : g_comms) {
+    ncclResult_t st = ncclSuccess;
+    ncclCommGetAsyncError(kv.second.comm, &st);
+    TORCH_CHECK(st == ncclSuccess, "RCCL communicator ", kv.first,
+                " failed asynchronously: ", ncclGetErrorString(st));
+  }
+}
+
+void comm_abort(int64_t id) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  auto it = g_comms.find(id);
+  if (it == g_comms.end()) return;
+  ncclCommAbort(it->second.comm);
+  hipFree(it->second.barrier_buf);
+  g_comms.erase(it);
+}
+
 void set_logging(bool enabled) { g_logging = enabled; }
 
 py::dict version_info() {
@@ -510,6 +534,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("comm_destroy", &comm_destroy);
   m.def("destroy_all_comms", &destroy_all_comms);
   m.def("comm_count", &comm_count);
+  m.def("check_async_errors", &check_async_errors);
+  m.def("comm_abort", &comm_abort);
   m.def("set_logging", &set_logging);
   m.def("version_info", &version_info);
   m.def("allreduce", &allreduce);

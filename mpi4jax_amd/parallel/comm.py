@@ -288,6 +288,14 @@ def flush():
     try:
         if torch.cuda.is_available() and torch.cuda.is_initialized():
             torch.cuda.synchronize()
+        from .._backend import rccl
+
+        if rccl.ext_is_loaded():
+            rccl.ext().check_async_errors()
+    except ImportError:
+        pass
+    except RuntimeError:
+        raise
     except Exception:
         pass
 
