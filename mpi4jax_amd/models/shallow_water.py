@@ -272,7 +272,10 @@ class ShallowWater:
             s1, s6, s7 = None, 8, 7
         elif (self.dtype == torch.float32
                 and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
-            s1, s6, s7 = 11, 16, 17
+            if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
+                s1, s6, s7 = 11, 16, 17
+            else:
+                s1, s6, s7 = None, 18, 17  # merged+vectorized single pass
         else:
             s1, s6, s7 = 1, 6, 7
         if s1 is not None:
