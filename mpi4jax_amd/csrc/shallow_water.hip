@@ -634,13 +634,8 @@ __global__ void sw_stage6v(SwArgs<float> a) {
 __global__ void sw_stage7v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int pp2 = (ppr + 1) / 2;  // two packs per thread (ILP)
-  const int t2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
-  if (t2 >= pp2 * ny) return;
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-  const int t = (t2 / pp2) * ppr + (t2 % pp2) * 2 + half;
-  if (t >= ppr * ny || t / ppr != t2 / pp2) break;
+  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
   const int jmin = a.f.south_open ? 1 : 2;
@@ -651,7 +646,7 @@ __global__ void sw_stage7v(SwArgs<float> a) {
                     i0 + 4 <= imax && i0 + 5 <= nx;
   if (!fast) {
     for (int c = 0; c < 4 && i0 + c < nx; ++c) stage7_cell(a, j, i0 + c);
-    continue;
+    return;
   }
   const long long idx = (long long)j * nx + i0;
   vf4 uc = ld4(a.u, idx), uw = ld4(a.u, idx - 1), ue = ld4(a.u, idx + 1);
@@ -665,7 +660,6 @@ __global__ void sw_stage7v(SwArgs<float> a) {
            (nu * (vn - vcc) / a.dy - nu * (vcc - vs) / a.dy) / a.dy;
   st4(a.u2, idx, uc + a.dt * lu);
   st4(a.v2, idx, vcc + a.dt * lv);
-  }
 }
 
 
@@ -697,13 +691,8 @@ __device__ inline vf4 shm1f(float xm1, vf4 A0) {
 __global__ void sw_stage18v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int pp2 = (ppr + 1) / 2;  // two packs per thread (ILP)
-  const int t2 = (int)(blockIdx.x * blockDim.x + threadIdx.x);
-  if (t2 >= pp2 * ny) return;
-#pragma unroll
-  for (int half = 0; half < 2; ++half) {
-  const int t = (t2 / pp2) * ppr + (t2 % pp2) * 2 + half;
-  if (t >= ppr * ny || t / ppr != t2 / pp2) break;
+  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
   const int jmin = a.f.south_open ? 1 : 2;
@@ -713,7 +702,7 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   const bool fast = j >= jmin && j <= jmax && i0 >= imin && i0 + 4 <= imax;
   if (!fast) {
     for (int c = 0; c < 4 && i0 + c < nx; ++c) stage8_cell(a, j, i0 + c);
-    continue;
+    return;
   }
   const long long idx = (long long)j * nx + i0;
   const long long idn = idx - nx, idp = idx + nx;
@@ -793,7 +782,6 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   st4(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
   st4(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
   st4(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
-  }
 }
 
 int sw_grid(long long n) {
@@ -867,9 +855,6 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
     sw_fill_args(a, p);
     long long ppr = (p.nx + 3) / 4;
     long long packs = ppr * p.ny;
-    if (stage == 17 || stage == 18) {  // ILP2 kernels: 2 packs/thread
-      packs = ((ppr + 1) / 2) * p.ny;
-    }
     dim3 grid((unsigned)((packs + 255) / 256)), block(256);
     switch (stage) {
       case 11: hipLaunchKernelGGL(sw_stage1v, grid, block, 0, stream, a); break;
