@@ -49,6 +49,7 @@ from .parallel.comm import (  # noqa: F401
 from ._backend import has_rccl_support, has_cuda_support  # noqa: F401
 from .utils.logging import set_logging, get_logging  # noqa: F401
 
+from .ops import jit_ops  # noqa: F401
 from .ops import (  # noqa: F401
     allgather,
     allreduce,
@@ -98,6 +99,7 @@ __all__ = [
     "finalize",
     "get_default_comm",
     "COMM_WORLD",
+    "jit_ops",
     "set_logging",
     "get_logging",
     "NOTSET",

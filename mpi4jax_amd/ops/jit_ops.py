@@ -1,0 +1,208 @@
+"""torch.compile-traceable collectives.
+
+The reference's headline capability is communication *inside* ``jax.jit``.
+This framework's equivalents are (a) hipGraph capture of eager regions
+(see ``ShallowWater.make_stepper``) and (b) these ``torch.library`` custom
+ops, which trace into ``torch.compile`` graphs with ``fullgraph=True`` —
+no graph breaks — including autograd (allreduce-SUM identity VJP and the
+sendrecv source/dest swap, same rules as the eager ops).
+
+Usage::
+
+    from mpi4jax_amd import jit_ops
+
+    @torch.compile(fullgraph=True)
+    def step(x):
+        return jit_ops.allreduce(x, "sum").sum()
+
+Communicators cross the graph boundary as int registry keys (the same
+int64-handle marshalling idea the reference uses for MPI objects,
+``mpi_ops_common.h:36-48``).
+"""
+
+import torch
+
+from ..parallel.comm import Communicator, resolve_comm
+from .reduce_ops import Op
+from .._backend import backend_for
+
+# int key <-> Communicator registry (keys are stable for a process life)
+_COMMS = {}
+_KEYS = {}
+
+
+@torch._dynamo.assume_constant_result
+def comm_key(comm) -> int:
+    """Registry key for a communicator.
+
+    Marked constant for dynamo: at trace time it runs eagerly (communicator
+    creation is not traceable) and the key is baked into the graph — dynamo
+    guards on the ``comm`` argument, so a different communicator object
+    triggers a recompile with its own key.
+    """
+    comm = resolve_comm(comm)
+    if id(comm) not in _KEYS:
+        key = len(_COMMS) + 1
+        _COMMS[key] = comm
+        _KEYS[id(comm)] = key
+    return _KEYS[id(comm)]
+
+
+def _comm(key: int) -> Communicator:
+    return _COMMS[key]
+
+
+_OPS = {o.value: o for o in Op}
+
+
+# ----------------------------------------------------------------- allreduce
+@torch.library.custom_op("mpi4jax_amd::allreduce", mutates_args=())
+def _allreduce(x: torch.Tensor, op: str, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).allreduce(x.contiguous(), _OPS[op], comm)
+
+
+@_allreduce.register_fake
+def _(x, op, key):
+    return torch.empty_like(x)
+
+
+def _allreduce_setup(ctx, inputs, output):
+    ctx.op = inputs[1]
+
+
+def _allreduce_bwd(ctx, grad):
+    if ctx.op != "sum":
+        raise RuntimeError("allreduce is only differentiable for op=sum")
+    return grad, None, None  # identity VJP (reference allreduce.py:152-159)
+
+
+torch.library.register_autograd("mpi4jax_amd::allreduce", _allreduce_bwd,
+                                setup_context=_allreduce_setup)
+
+
+def allreduce(x, op=Op.SUM, *, comm=None):
+    op = op.value if isinstance(op, Op) else str(op)
+    return _allreduce(x, op, comm_key(comm))
+
+
+# ----------------------------------------------------------------- allgather
+@torch.library.custom_op("mpi4jax_amd::allgather", mutates_args=())
+def _allgather(x: torch.Tensor, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).allgather(x.contiguous(), comm)
+
+
+@_allgather.register_fake
+def _(x, key):
+    size = _comm(key).size
+    return x.new_empty((size,) + tuple(x.shape))
+
+
+def allgather(x, *, comm=None):
+    return _allgather(x, comm_key(comm))
+
+
+# ----------------------------------------------------------------- alltoall
+@torch.library.custom_op("mpi4jax_amd::alltoall", mutates_args=())
+def _alltoall(x: torch.Tensor, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).alltoall(x.contiguous(), comm)
+
+
+@_alltoall.register_fake
+def _(x, key):
+    return torch.empty_like(x)
+
+
+def alltoall(x, *, comm=None):
+    return _alltoall(x, comm_key(comm))
+
+
+# ------------------------------------------------------------ reduce_scatter
+@torch.library.custom_op("mpi4jax_amd::reduce_scatter", mutates_args=())
+def _reduce_scatter(x: torch.Tensor, op: str, key: int) -> torch.Tensor:
+    from .reduce_scatter import reduce_scatter as rs
+
+    return rs(x, _OPS[op], comm=_comm(key))
+
+
+@_reduce_scatter.register_fake
+def _(x, op, key):
+    return x.new_empty(tuple(x.shape[1:]))
+
+
+def reduce_scatter(x, op=Op.SUM, *, comm=None):
+    op = op.value if isinstance(op, Op) else str(op)
+    return _reduce_scatter(x, op, comm_key(comm))
+
+
+# ----------------------------------------------------------------- bcast
+@torch.library.custom_op("mpi4jax_amd::bcast", mutates_args=())
+def _bcast(x: torch.Tensor, root: int, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).bcast(x.contiguous(), root, comm)
+
+
+@_bcast.register_fake
+def _(x, root, key):
+    return torch.empty_like(x)
+
+
+def bcast(x, root, *, comm=None):
+    return _bcast(x, root, comm_key(comm))
+
+
+# ----------------------------------------------------------------- scan
+@torch.library.custom_op("mpi4jax_amd::scan", mutates_args=())
+def _scan(x: torch.Tensor, op: str, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).scan(x.contiguous(), _OPS[op], comm)
+
+
+@_scan.register_fake
+def _(x, op, key):
+    return torch.empty_like(x)
+
+
+def scan(x, op=Op.SUM, *, comm=None):
+    op = op.value if isinstance(op, Op) else str(op)
+    return _scan(x, op, comm_key(comm))
+
+
+# ----------------------------------------------------------------- sendrecv
+@torch.library.custom_op("mpi4jax_amd::sendrecv", mutates_args=())
+def _sendrecv(sendbuf: torch.Tensor, recvbuf: torch.Tensor, source: int,
+              dest: int, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(sendbuf).sendrecv(
+        sendbuf.contiguous(), recvbuf, source, dest, 0, -1, comm, None
+    )
+
+
+@_sendrecv.register_fake
+def _(sendbuf, recvbuf, source, dest, key):
+    return torch.empty_like(recvbuf)
+
+
+def _sendrecv_setup(ctx, inputs, output):
+    _, recvbuf, source, dest, key = inputs
+    ctx.meta = (source, dest, key)
+    ctx.send_shape = tuple(inputs[0].shape)
+
+
+def _sendrecv_bwd(ctx, grad):
+    source, dest, key = ctx.meta
+    template = grad.new_empty(ctx.send_shape)
+    # VJP routes the cotangent along the reversed edge
+    # (reference sendrecv.py:278-293)
+    g = _sendrecv(grad.contiguous(), template, dest, source, key)
+    return g, None, None, None, None
+
+
+torch.library.register_autograd("mpi4jax_amd::sendrecv", _sendrecv_bwd,
+                                setup_context=_sendrecv_setup)
+
+
+def sendrecv(sendbuf, recvbuf, source, dest, *, comm=None):
+    return _sendrecv(sendbuf, recvbuf, source, dest, comm_key(comm))

@@ -181,6 +181,8 @@ def init(device=None):
     user-visible behavior matches.
     """
     global _WORLD
+    if _WORLD is not None:  # lock-free fast path (also torch.compile-safe)
+        return _WORLD
     with _LOCK:
         if _WORLD is not None:
             return _WORLD
@@ -244,6 +246,8 @@ def get_default_comm():
     isolates this library's traffic from any other torch.distributed use.
     """
     global _DEFAULT_COMM
+    if _DEFAULT_COMM is not None:  # lock-free fast path (compile-safe)
+        return _DEFAULT_COMM
     with _LOCK:
         if _DEFAULT_COMM is None:
             _DEFAULT_COMM = init().Clone()

@@ -1,0 +1,81 @@
+"""torch.compile integration: collectives trace with fullgraph=True."""
+
+import pytest
+import torch
+
+import mpi4jax_amd as m
+from mpi4jax_amd import jit_ops
+
+
+def test_allreduce_compiles_fullgraph():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allreduce(x, "sum") * 2
+
+    x = torch.arange(4.0)
+    y = f(x)
+    assert torch.equal(y, x * 2)
+
+
+def test_allreduce_compiled_grad():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allreduce(x, "sum").sum()
+
+    x = torch.randn(5, requires_grad=True)
+    f(x).backward()
+    assert torch.equal(x.grad, torch.ones(5))
+
+
+def test_allgather_compiled_shape():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allgather(x)
+
+    y = f(torch.zeros(3, 2))
+    assert y.shape == (1, 3, 2)
+
+
+def test_sendrecv_compiled_self_and_grad():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.sendrecv(x, x.detach(), source=0, dest=0).sum()
+
+    x = torch.randn(4, requires_grad=True)
+    f(x).backward()
+    assert torch.equal(x.grad, torch.ones(4))
+
+
+def test_scan_alltoall_bcast_compiled():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        a = jit_ops.scan(x, "sum")
+        b = jit_ops.alltoall(x[None])[0]
+        c = jit_ops.bcast(x, 0)
+        d = jit_ops.reduce_scatter(x[None], "sum")
+        return a + b + c + d
+
+    x = torch.arange(6.0)
+    assert torch.equal(f(x), 4 * x)
+
+
+def test_nonsum_compiled_grad_raises():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allreduce(x, "max").sum()
+
+    x = torch.randn(3, requires_grad=True)
+    with pytest.raises(RuntimeError, match="only differentiable"):
+        f(x).backward()
+
+
+@pytest.mark.gpu
+def test_compiled_allreduce_gpu():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allreduce(x, "sum") + 1
+
+    x = torch.arange(8.0, device="cuda")
+    y = f(x)
+    torch.cuda.synchronize()
+    assert torch.equal(y, x + 1)
