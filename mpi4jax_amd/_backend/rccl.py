@@ -22,7 +22,6 @@ Dtype strategy:
 import torch
 
 from ..ops.reduce_ops import Op, RCCL_OP_ENUM
-from ..utils.dtypes import COMPLEX_AS_REAL
 from ..utils.status import ANY_SOURCE, ANY_TAG
 from ..utils.logging import debug_timer, get_logging
 

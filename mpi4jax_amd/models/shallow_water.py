@@ -23,7 +23,7 @@ from collections import namedtuple
 import torch
 
 from ..parallel.comm import resolve_comm
-from ..parallel.grid import CartesianGrid, default_dims
+from ..parallel.grid import CartesianGrid
 
 ModelState = namedtuple("ModelState", "h u v dh du dv")
 

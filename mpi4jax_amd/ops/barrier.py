@@ -10,7 +10,6 @@ proceeds past it.  For a host-visible barrier, follow with
 """
 
 from ..parallel.comm import resolve_comm
-from .._backend import backend_for
 from ..utils.tokens import NOTSET, raise_if_token_is_set
 
 import torch

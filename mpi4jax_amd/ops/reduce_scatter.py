@@ -10,7 +10,7 @@ r across all ranks.
 import torch
 
 from ..utils.tokens import NOTSET, raise_if_token_is_set
-from .reduce_ops import Op, resolve_op
+from .reduce_ops import resolve_op
 from ._common import prepare
 
 

@@ -11,8 +11,6 @@ On the RCCL backend each direction's sendrecv is grouped, so a full
 on the compute stream with zero host synchronization.
 """
 
-import torch
-
 from .comm import resolve_comm
 
 
