@@ -90,10 +90,11 @@ def main():
     # warmup (untimed) + hipGraph capture of a 2-step multistep
     for _ in range(args.warmup):
         state = sw.step(state)
-    advance, state = sw.make_stepper(state, steps_per_call=2)
+    spc = 10 if args.steps >= 50 else 2
+    advance, state = sw.make_stepper(state, steps_per_call=spc)
     advance()  # one warm replay
 
-    n_calls, rem = divmod(args.steps, 2)
+    n_calls, rem = divmod(args.steps, spc)
     m.barrier()
     if use_gpu:
         torch.cuda.synchronize()

@@ -46,7 +46,11 @@ from .parallel.comm import (  # noqa: F401
     get_world,
     COMM_WORLD,
 )
-from ._backend import has_rccl_support, has_cuda_support  # noqa: F401
+from ._backend import (  # noqa: F401
+    has_rccl_support,
+    has_cuda_support,
+    has_sycl_support,
+)
 from .utils.logging import set_logging, get_logging  # noqa: F401
 
 from .ops import jit_ops  # noqa: F401
@@ -82,6 +86,7 @@ __all__ = [
     "sendrecv",
     "has_rccl_support",
     "has_cuda_support",
+    "has_sycl_support",
     "Op",
     "SUM",
     "PROD",

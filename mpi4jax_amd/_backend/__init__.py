@@ -40,3 +40,9 @@ def has_rccl_support() -> bool:
 # means the RCCL extension.
 def has_cuda_support() -> bool:
     return has_rccl_support()
+
+
+def has_sycl_support() -> bool:
+    """Always False — single-vendor by design (drop-in parity with the
+    reference's capability probe, utils.py:159-174)."""
+    return False
