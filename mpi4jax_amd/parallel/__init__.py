@@ -8,3 +8,4 @@ from .comm import (  # noqa: F401
     COMM_WORLD,
 )
 from .grid import CartesianGrid  # noqa: F401
+from .ddp import average_gradients  # noqa: F401
