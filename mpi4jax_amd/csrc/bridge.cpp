@@ -437,6 +437,9 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
   bool is_double = h.scalar_type() == at::kDouble;
   TORCH_CHECK(is_double || h.scalar_type() == at::kFloat,
               "shallow-water kernels support f32/f64");
+  TORCH_CHECK(h.numel() < (int64_t)1 << 31,
+              "shallow-water kernels index with 32-bit math; local domain "
+              "must have fewer than 2^31 cells");
   SwLaunchParams p;
   void** slots[16] = {&p.fe, &p.fn, &p.q, &p.ke, &p.h, &p.u, &p.v,
                       &p.dnh, &p.dnu, &p.dnv, &p.doh, &p.dou, &p.dov,

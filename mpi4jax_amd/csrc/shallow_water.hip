@@ -559,7 +559,15 @@ __device__ inline void stage7_cell(const SwArgs<float>& a, int j, int i) {
 __global__ void sw_stage1v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  // XCD-aware block remap (same rationale as SW_BLOCK_MAP): give each XCD
+  // a contiguous band of rows so j±1 stencil rows hit that XCD's L2
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
   if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
@@ -591,7 +599,15 @@ __global__ void sw_stage1v(SwArgs<float> a) {
 __global__ void sw_stage6v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  // XCD-aware block remap (same rationale as SW_BLOCK_MAP): give each XCD
+  // a contiguous band of rows so j±1 stencil rows hit that XCD's L2
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
   if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
@@ -634,7 +650,15 @@ __global__ void sw_stage6v(SwArgs<float> a) {
 __global__ void sw_stage7v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  // XCD-aware block remap (same rationale as SW_BLOCK_MAP): give each XCD
+  // a contiguous band of rows so j±1 stencil rows hit that XCD's L2
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
   if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
@@ -691,7 +715,15 @@ __device__ inline vf4 shm1f(float xm1, vf4 A0) {
 __global__ void sw_stage18v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
-  const int t = (int)(blockIdx.x * blockDim.x + threadIdx.x);
+  // XCD-aware block remap (same rationale as SW_BLOCK_MAP): give each XCD
+  // a contiguous band of rows so j±1 stencil rows hit that XCD's L2
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
   if (t >= ppr * ny) return;
   const int j = t / ppr;
   const int i0 = (t % ppr) * 4;
