@@ -99,6 +99,8 @@ def _reduction_view(x, op, op_name):
 
 def allreduce(x, op, comm):
     _check_op(op, x, "allreduce")
+    if x.numel() == 0:
+        return x.clone()
     with debug_timer("Allreduce", comm.rank, f"{x.numel()} items"):
         xr, rop, post = _reduction_view(x.contiguous(), op, "allreduce")
         out = torch.empty_like(xr)
@@ -127,6 +129,8 @@ def reduce(x, op, root, comm):
 
 def scan(x, op, comm):
     _check_op(op, x, "scan")
+    if x.numel() == 0:
+        return x.clone()
     if op is Op.AVG:
         raise ValueError("scan: AVG is not a valid scan operator")
     with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
