@@ -19,6 +19,8 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
+#include <roctracer/roctx.h>
+
 #include <cstdio>
 #include <mutex>
 #include <string>
@@ -119,6 +121,14 @@ void log_enqueue(const char* op, const CommEntry& c, int64_t items) {
   }
 }
 
+// roctx range around every collective enqueue — shows up in
+// `rocprofv3 --marker-trace` timelines (tracing spans)
+struct RoctxRange {
+  explicit RoctxRange(const char* name) { roctxRangePushA(name); }
+  ~RoctxRange() { roctxRangePop(); }
+};
+#define ROCTX_SCOPE(name) RoctxRange roctx_scope_(name)
+
 // ---------------------------------------------------------------- lifecycle
 
 py::bytes get_unique_id() {
@@ -214,6 +224,7 @@ py::dict version_info() {
 // --------------------------------------------------------------- collectives
 
 void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::allreduce");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(out.numel() == in.numel(), "size mismatch");
@@ -225,6 +236,7 @@ void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
 
 void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
             int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::reduce");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   log_enqueue("Reduce", c, in.numel());
@@ -234,6 +246,7 @@ void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
 }
 
 void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::allgather");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(out.numel() == in.numel() * c.size, "allgather size mismatch");
@@ -243,6 +256,7 @@ void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
 }
 
 void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::bcast");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   log_enqueue("Bcast", c, in.numel());
@@ -252,6 +266,7 @@ void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 
 void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
                     int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::reduce_scatter");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(in.numel() == out.numel() * c.size,
@@ -268,6 +283,7 @@ void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
 // 153 GB/s link; no forwarding needed).
 
 void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::alltoall");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(in.numel() % c.size == 0, "alltoall count not divisible");
@@ -288,6 +304,7 @@ void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
 }
 
 void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::gather");
   auto& c = get_comm(comm_id);
   TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad gather input");
   int64_t chunk = in.numel();
@@ -309,6 +326,7 @@ void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 }
 
 void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::scatter");
   auto& c = get_comm(comm_id);
   TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad scatter output");
   int64_t chunk = out.numel();
@@ -330,6 +348,7 @@ void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 }
 
 void send(at::Tensor in, int64_t dest, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::send");
   auto& c = get_comm(comm_id);
   TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad send input");
   log_enqueue("Send", c, in.numel());
@@ -338,6 +357,7 @@ void send(at::Tensor in, int64_t dest, int64_t comm_id) {
 }
 
 void recv(at::Tensor out, int64_t source, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::recv");
   auto& c = get_comm(comm_id);
   TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad recv output");
   log_enqueue("Recv", c, out.numel());
@@ -347,6 +367,7 @@ void recv(at::Tensor out, int64_t source, int64_t comm_id) {
 
 void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
               int64_t dest, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::sendrecv");
   auto& c = get_comm(comm_id);
   TORCH_CHECK(sendbuf.is_cuda() && sendbuf.is_contiguous(), "bad sendbuf");
   TORCH_CHECK(recvbuf.is_cuda() && recvbuf.is_contiguous(), "bad recvbuf");
@@ -362,6 +383,7 @@ void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
 }
 
 void barrier(int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::barrier");
   auto& c = get_comm(comm_id);
   log_enqueue("Barrier", c, 1);
   // tiny allreduce on persistent scratch = a cross-rank stream barrier
@@ -375,6 +397,7 @@ void barrier(int64_t comm_id) {
 // the CDNA4 kernel, forward to r+1.  All stream-ordered; the host never
 // blocks.
 void scan(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::scan");
   auto& c = get_comm(comm_id);
   check_pair(out, in);
   auto dt = nccl_dtype(in);
