@@ -60,7 +60,8 @@ std::unordered_map<int64_t, CommEntry> g_comms;
 int64_t g_next_id = 1;
 bool g_logging = false;
 
-CommEntry& get_comm(int64_t id) {
+CommEntry get_comm(int64_t id) {  // by value: entries are tiny PODs and a
+  // reference would dangle if another thread destroyed the comm
   std::lock_guard<std::mutex> lk(g_mutex);
   auto it = g_comms.find(id);
   TORCH_CHECK(it != g_comms.end(), "unknown RCCL communicator handle ", id);
@@ -225,7 +226,7 @@ py::dict version_info() {
 
 void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::allreduce");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(out.numel() == in.numel(), "size mismatch");
   log_enqueue("Allreduce", c, in.numel());
@@ -237,7 +238,7 @@ void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
 void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
             int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::reduce");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   log_enqueue("Reduce", c, in.numel());
   RCCL_CHECK(ncclReduce(in.data_ptr(), out.data_ptr(), in.numel(),
@@ -247,7 +248,7 @@ void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
 
 void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::allgather");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(out.numel() == in.numel() * c.size, "allgather size mismatch");
   log_enqueue("Allgather", c, in.numel());
@@ -257,7 +258,7 @@ void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
 
 void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::bcast");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   log_enqueue("Bcast", c, in.numel());
   RCCL_CHECK(ncclBroadcast(in.data_ptr(), out.data_ptr(), in.numel(),
@@ -267,7 +268,7 @@ void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
                     int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::reduce_scatter");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(in.numel() == out.numel() * c.size,
               "reduce_scatter size mismatch");
@@ -284,7 +285,7 @@ void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
 
 void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::alltoall");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   TORCH_CHECK(in.numel() % c.size == 0, "alltoall count not divisible");
   int64_t chunk = in.numel() / c.size;
@@ -305,7 +306,7 @@ void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
 
 void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::gather");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad gather input");
   int64_t chunk = in.numel();
   int64_t esz = in.element_size();
@@ -327,7 +328,7 @@ void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 
 void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::scatter");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad scatter output");
   int64_t chunk = out.numel();
   int64_t esz = out.element_size();
@@ -349,7 +350,7 @@ void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
 
 void send(at::Tensor in, int64_t dest, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::send");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad send input");
   log_enqueue("Send", c, in.numel());
   RCCL_CHECK(ncclSend(in.data_ptr(), in.numel(), nccl_dtype(in), (int)dest,
@@ -358,7 +359,7 @@ void send(at::Tensor in, int64_t dest, int64_t comm_id) {
 
 void recv(at::Tensor out, int64_t source, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::recv");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad recv output");
   log_enqueue("Recv", c, out.numel());
   RCCL_CHECK(ncclRecv(out.data_ptr(), out.numel(), nccl_dtype(out),
@@ -368,7 +369,7 @@ void recv(at::Tensor out, int64_t source, int64_t comm_id) {
 void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
               int64_t dest, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::sendrecv");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   TORCH_CHECK(sendbuf.is_cuda() && sendbuf.is_contiguous(), "bad sendbuf");
   TORCH_CHECK(recvbuf.is_cuda() && recvbuf.is_contiguous(), "bad recvbuf");
   log_enqueue("Sendrecv", c, sendbuf.numel());
@@ -384,7 +385,7 @@ void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
 
 void barrier(int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::barrier");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   log_enqueue("Barrier", c, 1);
   // tiny allreduce on persistent scratch = a cross-rank stream barrier
   // 
@@ -398,7 +399,7 @@ void barrier(int64_t comm_id) {
 // blocks.
 void scan(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
   ROCTX_SCOPE("mpi4jax_amd::scan");
-  auto& c = get_comm(comm_id);
+  auto c = get_comm(comm_id);
   check_pair(out, in);
   auto dt = nccl_dtype(in);
   hipStream_t stream = cur_stream();
