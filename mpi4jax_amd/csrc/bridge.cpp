@@ -19,7 +19,7 @@
 #include <hip/hip_runtime.h>
 #include <rccl/rccl.h>
 
-#include <roctracer/roctx.h>
+#include <rocprofiler-sdk-roctx/roctx.h>
 
 #include <cstdio>
 #include <mutex>

@@ -94,7 +94,7 @@ def build_native(force=False):
             "-lc10",
             "-lc10_hip",
             "-lrccl",
-            "-lroctx64",
+            "-lrocprofiler-sdk-roctx",
             "-lamdhip64",
         ]
         _run(link)
