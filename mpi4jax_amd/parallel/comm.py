@@ -88,21 +88,16 @@ class Communicator:
     def Split(self, color, key=0):
         """Split into sub-communicators by color, ordered by (key, rank).
 
-        Collective over this communicator.
+        Collective over this communicator (group creation uses local
+        synchronization, so only this communicator's members take part —
+        splitting a sub-communicator works).
         """
         info = self._allgather_py((int(color), int(key)))
         groups = {}
         for comm_rank, (c, k) in enumerate(info):
             groups.setdefault(c, []).append((k, comm_rank))
-        # deterministic iteration: every process must create every group
-        my_comm = None
-        for c in sorted(groups):
-            members = [self._ranks[r] for _, r in sorted(groups[c])]
-            comm = _new_comm(members, f"{self._label}.split{c}",
-                             participate=self._world_rank in members)
-            if c == int(color):
-                my_comm = comm
-        return my_comm
+        members = [self._ranks[r] for _, r in sorted(groups[int(color)])]
+        return _new_comm(members, f"{self._label}.split{int(color)}")
 
     def free(self):
         """Release the native RCCL communicator (gloo groups are pooled)."""
@@ -214,8 +209,12 @@ def init(device=None):
         return _WORLD
 
 
-def _new_comm(ranks, label, participate=True):
-    """Create a communicator over the given global ranks (collective)."""
+def _new_comm(ranks, label):
+    """Create a communicator over the given global ranks.
+
+    Collective over ``ranks`` only (``use_local_synchronization``), so
+    sub-communicators can be cloned/split without involving the world.
+    """
     world = init()
     if world.size == 1:
         _COMM_COUNTER[0] += 1
@@ -223,10 +222,8 @@ def _new_comm(ranks, label, participate=True):
             ranks=[0], gloo_group=None, parent_world_rank=0,
             label=f"{label}#{_COMM_COUNTER[0]}",
         )
-    group = dist.new_group(ranks)  # collective over the world
+    group = dist.new_group(ranks, use_local_synchronization=True)
     _COMM_COUNTER[0] += 1
-    if not participate:
-        return None
     return Communicator(
         ranks=ranks,
         gloo_group=group,

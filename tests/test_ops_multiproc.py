@@ -204,3 +204,19 @@ def _reduce_scatter(rank, ws):
 
 def test_reduce_scatter_multiproc():
     run_multiproc(_reduce_scatter, 2)
+
+
+def _nested_split(rank, ws):
+    # Split of a Split must not hang (group creation is local-sync)
+    world = m.get_world()
+    sub = world.Split(color=0, key=rank)  # everyone
+    sub2 = sub.Split(color=sub.rank % 2, key=0)  # split the sub-comm
+    x = torch.tensor([1.0])
+    total = m.allreduce(x, m.SUM, comm=sub2).item()
+    assert total == sub2.size
+    clone = sub2.Clone()
+    assert m.allreduce(x, m.SUM, comm=clone).item() == sub2.size
+
+
+def test_nested_split():
+    run_multiproc(_nested_split, 2)
