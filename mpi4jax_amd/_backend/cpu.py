@@ -23,6 +23,13 @@ def _self_queue(comm):
     return _SELF_QUEUES.setdefault(id(comm), [])
 
 
+def _wire(t):
+    """1-D uint8 view sharing storage — gloo supports every op on bytes,
+    so exotic dtypes (bf16/f16/int16/complex/bool) travel unchanged."""
+    return t.view(torch.uint8) if t.dtype == torch.uint8 else \
+        t.contiguous().flatten().view(torch.uint8)
+
+
 def _gloo_reduce_op(op: Op, dtype):
     if op is Op.AVG:
         # gloo has no AVG; emulate with SUM + divide
@@ -73,9 +80,14 @@ def allgather(x, comm):
         xc = x.contiguous()
         if comm.size == 1:
             return xc[None].clone()
-        out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
-        dist.all_gather(list(out.unbind(0)), xc, group=comm.gloo_group)
-        return out
+        if xc.numel() == 0:
+            return torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
+        nbytes = xc.numel() * xc.element_size()
+        ob = torch.empty((comm.size, max(nbytes, 1)), dtype=torch.uint8)
+        dist.all_gather(list(ob.unbind(0)), _wire(xc),
+                        group=comm.gloo_group)
+        return ob.flatten().view(x.dtype).reshape(
+            (comm.size,) + tuple(x.shape))
 
 
 def alltoall(x, comm):
@@ -84,7 +96,7 @@ def alltoall(x, comm):
         if comm.size == 1:
             return xc.clone()
         out = torch.empty_like(xc)
-        dist.all_to_all_single(out, xc, group=comm.gloo_group)
+        dist.all_to_all_single(_wire(out), _wire(xc), group=comm.gloo_group)
         return out
 
 
@@ -98,7 +110,8 @@ def bcast(x, root, comm):
     with debug_timer("Bcast", comm.rank, f"{x.numel()} items"):
         out = x.clone().contiguous()
         if comm.size > 1:
-            dist.broadcast(out, src=comm.global_rank(root), group=comm.gloo_group)
+            dist.broadcast(_wire(out), src=comm.global_rank(root),
+                           group=comm.gloo_group)
         return out
 
 
@@ -107,14 +120,20 @@ def gather(x, root, comm):
         xc = x.contiguous()
         if comm.size == 1:
             return xc[None].clone() if comm.rank == root else None
+        if xc.numel() == 0:
+            return (torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
+                    if comm.rank == root else None)
         if comm.rank == root:
-            out = torch.empty((comm.size,) + tuple(x.shape), dtype=x.dtype)
+            nbytes = xc.numel() * xc.element_size()
+            ob = torch.empty((comm.size, max(nbytes, 1)), dtype=torch.uint8)
             dist.gather(
-                xc, list(out.unbind(0)), dst=comm.global_rank(root),
+                _wire(xc), list(ob.unbind(0)), dst=comm.global_rank(root),
                 group=comm.gloo_group,
             )
-            return out
-        dist.gather(xc, None, dst=comm.global_rank(root), group=comm.gloo_group)
+            return ob.flatten().view(x.dtype).reshape(
+                (comm.size,) + tuple(x.shape))
+        dist.gather(_wire(xc), None, dst=comm.global_rank(root),
+                    group=comm.gloo_group)
         return None
 
 
@@ -126,13 +145,14 @@ def scatter(x, root, comm):
             xc = x.contiguous()
             out = torch.empty(tuple(x.shape[1:]), dtype=x.dtype)
             dist.scatter(
-                out, [t.contiguous() for t in xc.unbind(0)],
+                _wire(out), [_wire(t.contiguous()) for t in xc.unbind(0)],
                 src=comm.global_rank(root), group=comm.gloo_group,
             )
         else:
             out = torch.empty(tuple(x.shape), dtype=x.dtype)
             dist.scatter(
-                out, None, src=comm.global_rank(root), group=comm.gloo_group
+                _wire(out), None, src=comm.global_rank(root),
+                group=comm.gloo_group
             )
         return out
 
@@ -149,11 +169,12 @@ def scan(x, op, comm):
             return out
         if comm.rank > 0:
             partial = torch.empty_like(out)
-            dist.recv(partial, src=comm.global_rank(comm.rank - 1),
+            dist.recv(_wire(partial), src=comm.global_rank(comm.rank - 1),
                       group=comm.gloo_group, tag=771)
             out = combine(op, partial, out)
         if comm.rank < comm.size - 1:
-            dist.send(out.contiguous(), dst=comm.global_rank(comm.rank + 1),
+            dist.send(_wire(out.contiguous()),
+                      dst=comm.global_rank(comm.rank + 1),
                       group=comm.gloo_group, tag=771)
         return out
 
@@ -164,8 +185,8 @@ def send(x, dest, tag, comm):
         if dest == comm.rank:
             _self_queue(comm).append((tag, xc))
             return
-        dist.send(xc, dst=comm.global_rank(dest), group=comm.gloo_group,
-                  tag=max(tag, 0))
+        dist.send(_wire(xc), dst=comm.global_rank(dest),
+                  group=comm.gloo_group, tag=max(tag, 0))
 
 
 def recv(template, source, tag, comm, status):
@@ -185,7 +206,7 @@ def recv(template, source, tag, comm, status):
                 "recv from self with no matching buffered send"
             )
         src = None if source == ANY_SOURCE else comm.global_rank(source)
-        sender = dist.recv(out, src=src, group=comm.gloo_group,
+        sender = dist.recv(_wire(out), src=src, group=comm.gloo_group,
                            tag=max(tag, 0))
         src_comm_rank = (
             source if source != ANY_SOURCE
@@ -205,19 +226,16 @@ def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
             _fill_status(status, source, recvtag, out)
             return out
         sc = sendbuf.contiguous()
-        # even ranks send first — deadlock-free pairing for blocking gloo p2p
-        first_send = comm.rank <= source if dest != comm.rank else False
-        ops = []
         sreq = dist.P2POp(
-            dist.isend, sc, peer=comm.global_rank(dest),
+            dist.isend, _wire(sc), peer=comm.global_rank(dest),
             group=comm.gloo_group, tag=max(sendtag, 0),
         )
         rreq = dist.P2POp(
-            dist.irecv, out, peer=comm.global_rank(source),
+            dist.irecv, _wire(out), peer=comm.global_rank(source),
             group=comm.gloo_group, tag=max(recvtag, 0),
         )
-        ops = [sreq, rreq] if not first_send else [rreq, sreq]
-        reqs = dist.batch_isend_irecv(ops)
+        # batched non-blocking pair: deadlock-free regardless of order
+        reqs = dist.batch_isend_irecv([sreq, rreq])
         for r in reqs:
             r.wait()
         _fill_status(status, source, recvtag, out)

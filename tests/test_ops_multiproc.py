@@ -220,3 +220,30 @@ def _nested_split(rank, ws):
 
 def test_nested_split():
     run_multiproc(_nested_split, 2)
+
+
+def _exotic_dtypes(rank, ws):
+    # movement ops across the wire for dtypes MPI/gloo treat specially
+    for dtype in (torch.bfloat16, torch.float16, torch.int16,
+                  torch.complex64, torch.bool):
+        if dtype == torch.bool:
+            x = torch.tensor([rank % 2 == 0, True, False])
+        elif dtype == torch.complex64:
+            x = (torch.arange(6.) + 1j * rank).to(dtype)
+        else:
+            x = (torch.arange(6.) + rank).to(dtype)
+        y = m.bcast(x.clone(), 0)
+        g = m.allgather(x)
+        assert g.dtype == dtype and g.shape[0] == ws
+        a2a = m.alltoall(x[:ws].reshape(ws, 1).contiguous())
+        assert a2a.dtype == dtype
+        got = m.gather(x, 0)
+        if rank == 0:
+            assert got.shape[0] == ws
+        if dtype not in (torch.bool, torch.complex64):
+            s = m.scan(x, m.SUM)
+            assert s.dtype == dtype
+
+
+def test_exotic_dtypes_over_wire():
+    run_multiproc(_exotic_dtypes, 2)
