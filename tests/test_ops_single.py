@@ -133,3 +133,12 @@ def test_reduce_scatter(x):
     assert torch.equal(y, x)
     with pytest.raises(ValueError):
         m.reduce_scatter(torch.zeros(3, 2), m.SUM)
+
+
+def test_batched_collective_idiom():
+    """The reference registers vmap batching rules; here a batched
+    collective IS the collective on the stacked tensor (PARITY.md)."""
+    xs = torch.randn(4, 3, 2)  # a "batch" of 4 inputs
+    stacked = m.allreduce(xs, m.SUM)
+    looped = torch.stack([m.allreduce(x, m.SUM) for x in xs])
+    assert torch.equal(stacked, looped)
