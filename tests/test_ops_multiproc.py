@@ -247,3 +247,14 @@ def _exotic_dtypes(rank, ws):
 
 def test_exotic_dtypes_over_wire():
     run_multiproc(_exotic_dtypes, 2)
+
+
+def _usage_global_sum(rank, ws):
+    # the docs/usage.md headline example at world size 4
+    a = torch.zeros(3, 3)
+    result = m.allreduce(a + rank, m.SUM)
+    assert torch.equal(result, torch.full((3, 3), float(sum(range(ws)))))
+
+
+def test_usage_example_world4():
+    run_multiproc(_usage_global_sum, 4)
