@@ -544,6 +544,28 @@ void unpack_cols(std::vector<at::Tensor> fields, at::Tensor buf,
                      is_double ? 1 : 0, cur_stream());
 }
 
+void pack_corners(at::Tensor buf, std::vector<at::Tensor> fields) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  TORCH_CHECK(buf.is_cuda() && buf.is_contiguous() &&
+                  buf.numel() >= 4 * (int64_t)fields.size(),
+              "bad corner buffer");
+  launch_pack_corners(buf.data_ptr(), ptrs, (int)fields.size(), ny, nx,
+                      is_double ? 1 : 0, cur_stream());
+}
+
+void unpack_corners(std::vector<at::Tensor> fields, at::Tensor buf,
+                    int64_t mask) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  launch_unpack_corners(ptrs, buf.data_ptr(), (int)fields.size(), ny, nx,
+                        (int)mask, is_double ? 1 : 0, cur_stream());
+}
+
 // direct access to the combine kernel (used by gpu numerics tests)
 void combine(at::Tensor dst, at::Tensor a, at::Tensor b, int64_t op) {
   check_pair(dst, a);
@@ -587,4 +609,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("halo_wrap", &halo_wrap);
   m.def("pack_cols", &pack_cols);
   m.def("unpack_cols", &unpack_cols);
+  m.def("pack_corners", &pack_corners);
+  m.def("unpack_corners", &unpack_corners);
 }

@@ -66,3 +66,11 @@ void launch_pack_cols(void* buf, void* const* fields, int nf, long long ny,
 void launch_unpack_cols(void* const* fields, const void* buf, int nf,
                         long long ny, long long nx, long long col,
                         int is_double, hipStream_t stream);
+
+// corner pack/unpack for the single-group halo exchange
+void launch_pack_corners(void* buf, void* const* fields, int nf,
+                         long long ny, long long nx, int is_double,
+                         hipStream_t stream);
+void launch_unpack_corners(void* const* fields, const void* buf, int nf,
+                           long long ny, long long nx, int mask,
+                           int is_double, hipStream_t stream);

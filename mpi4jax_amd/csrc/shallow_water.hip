@@ -1013,3 +1013,65 @@ void launch_unpack_cols(void* const* fields, const void* buf, int nf,
                        dim3(kBlock), 0, stream, a, (const float*)buf);
   }
 }
+
+// corner pack/unpack for the single-group halo exchange: buf layout is
+// [diag d][field f] with d: 0=to-SW/from-NE, 1=to-SE/from-NW,
+// 2=to-NW/from-SE, 3=to-NE/from-SW (matching parallel/grid.py halo_plan).
+namespace {
+
+template <typename T>
+__global__ void pack_corners_kernel(HaloArgs<T> a, T* buf) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= 4 * a.nf) return;
+  int d = t / a.nf, f = t % a.nf;
+  const int ny = a.ny, nx = a.nx;
+  int j = (d < 2) ? 1 : ny - 2;
+  int i = (d == 0 || d == 2) ? 1 : nx - 2;
+  buf[t] = halo_field(a, f)[(long long)j * nx + i];
+}
+
+template <typename T>
+__global__ void unpack_corners_kernel(HaloArgs<T> a, const T* buf,
+                                      int mask) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= 4 * a.nf) return;
+  int d = t / a.nf, f = t % a.nf;
+  if (!(mask & (1 << d))) return;
+  const int ny = a.ny, nx = a.nx;
+  // recv cells: d0 (ny-1,nx-1), d1 (ny-1,0), d2 (0,nx-1), d3 (0,0)
+  int j = (d < 2) ? ny - 1 : 0;
+  int i = (d == 0 || d == 2) ? nx - 1 : 0;
+  halo_field(a, f)[(long long)j * nx + i] = buf[t];
+}
+
+}  // namespace
+
+void launch_pack_corners(void* buf, void* const* fields, int nf,
+                         long long ny, long long nx, int is_double,
+                         hipStream_t stream) {
+  int n = 4 * nf;
+  if (is_double) {
+    auto a = make_halo_args<double>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(pack_corners_kernel<double>, dim3(1), dim3(64), 0,
+                       stream, a, (double*)buf);
+  } else {
+    auto a = make_halo_args<float>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(pack_corners_kernel<float>, dim3(1), dim3(64), 0,
+                       stream, a, (float*)buf);
+  }
+  (void)n;
+}
+
+void launch_unpack_corners(void* const* fields, const void* buf, int nf,
+                           long long ny, long long nx, int mask,
+                           int is_double, hipStream_t stream) {
+  if (is_double) {
+    auto a = make_halo_args<double>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(unpack_corners_kernel<double>, dim3(1), dim3(64), 0,
+                       stream, a, (const double*)buf, mask);
+  } else {
+    auto a = make_halo_args<float>(fields, nf, ny, nx, 0);
+    hipLaunchKernelGGL(unpack_corners_kernel<float>, dim3(1), dim3(64), 0,
+                       stream, a, (const float*)buf, mask);
+  }
+}

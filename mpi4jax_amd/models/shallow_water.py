@@ -23,7 +23,7 @@ from collections import namedtuple
 import torch
 
 from ..parallel.comm import resolve_comm
-from ..parallel.grid import CartesianGrid
+from ..parallel.grid import CartesianGrid, halo_plan
 
 ModelState = namedtuple("ModelState", "h u v dh du dv")
 
@@ -34,25 +34,6 @@ CORIOLIS_F = 2e-4
 CORIOLIS_BETA = 2e-11
 ADAMS_BASHFORTH_A = 1.5 + 0.1
 ADAMS_BASHFORTH_B = -(0.5 + 0.1)
-
-def halo_phase_schedule(nx, ny):
-    """The fused halo exchange's phase plan.
-
-    Mirrors :meth:`CartesianGrid.halo_exchange`: both column exchanges
-    first (independent of each other), then both row exchanges (full rows,
-    carrying the freshly-filled column-halo corners).  Each entry:
-    ``(is_cols_phase, send_dir, recv_dir, recv_index, send_index)`` where
-    indices are columns for cols phases and rows otherwise.  The two
-    phases of each half are batched into one RCCL group by the fused path.
-    Validated against the eager exchange in tests/test_shallow_water.py.
-    """
-    return (
-        (True, "west", "east", nx - 1, 1),   # recv east halo, send col 1
-        (True, "east", "west", 0, nx - 2),   # recv west halo, send nx-2
-        (False, "north", "south", 0, ny - 2),  # recv row 0, send row ny-2
-        (False, "south", "north", ny - 1, 1),  # recv row ny-1, send row 1
-    )
-
 
 _I = slice(1, -1)  # interior
 _L = slice(None, -2)  # shifted left/down
@@ -299,71 +280,86 @@ class ShallowWater:
     # field memory (rows are contiguous — true zero-copy).  Phase order
     # matches the eager path (W,N,E,S) so halo corners compose identically.
     def _exchange_fields(self, fields):
+        """One-group halo exchange (halo_plan is order-independent).
+
+        Self-wrap columns run as a kernel (no RCCL — keeps the world-1
+        path graph-capturable with zero comm init); everything remote —
+        packed columns, in-place interior rows, diagonal corners — goes
+        into a SINGLE RCCL group enqueue.
+        """
         from .._backend import rccl
 
         ext = rccl.ext()
         g = self.grid
         me = self.comm.rank
         nf = len(fields)
-        ny = self.ny_local
+        ny, nx = self.ny_local, self.nx_local
         fb = self._fb
         if "col_sbuf0" not in fb or fb["col_sbuf0"].numel() < 3 * ny:
             for k in ("col_sbuf0", "col_rbuf0", "col_sbuf1", "col_rbuf1"):
                 fb[k] = torch.empty(3 * ny, dtype=self.dtype,
                                     device=self.device)
-        sched = halo_phase_schedule(self.nx_local, ny)
+            fb["cor_sbuf"] = torch.empty(12, dtype=self.dtype,
+                                         device=self.device)
+            fb["cor_rbuf"] = torch.empty_like(fb["cor_sbuf"])
+        plan = halo_plan(nx, ny)
 
-        # cols half: both column phases in ONE RCCL group (self-wrap runs
-        # as a single kernel and never touches RCCL — keeps the world-1
-        # path graph-capturable with zero comm init)
-        pend = []
-        col_nbrs = [(g.neighbor(sdir), g.neighbor(rdir))
-                    for _, sdir, rdir, _, _ in sched[:2]]
-        if all(st == me and rf == me for st, rf in col_nbrs):
-            ext.halo_wrap(fields, 2)  # both periodic wraps, one kernel
-            col_nbrs = None
-        for k, (_, sdir, rdir, ridx, sidx) in enumerate(sched[:2]):
-            if col_nbrs is None:
-                break
-            st, rf = col_nbrs[k]
+        # resolve peers; pre-group kernels (col/corner packs, wraps)
+        col_ops, row_ops, cor_ops = [], [], []
+        for k, (_, sdir, rdir, ridx, sidx) in enumerate(plan[:2]):
+            st, rf = g.neighbor(sdir), g.neighbor(rdir)
             if st is None and rf is None:
                 continue
             if st == me and rf == me:
-                ext.halo_wrap(fields,
-                              0 if ridx == self.nx_local - 1 else 1)
+                ext.halo_wrap(fields, 0 if ridx == nx - 1 else 1)
                 continue
             sbuf = fb[f"col_sbuf{k}"][:nf * ny]
             rbuf = fb[f"col_rbuf{k}"][:nf * ny]
             if st is not None:
                 ext.pack_cols(sbuf, fields, sidx)
-            pend.append((st, rf, sbuf, rbuf, ridx))
-        if pend:
-            comm_id = self.comm.rccl_handle()
-            ext.group_start()
-            for st, rf, sbuf, rbuf, _ in pend:
-                if st is not None:
-                    ext.send(sbuf, st, comm_id)
+            col_ops.append((st, rf, sbuf, rbuf, ridx))
+        for (_, sdir, rdir, ridx, sidx) in plan[2:4]:
+            st, rf = g.neighbor(sdir), g.neighbor(rdir)
+            if st is not None or rf is not None:
+                row_ops.append((st, rf, ridx, sidx))
+        cor_mask = 0
+        for d, (_, sdiag, rdiag, _, _) in enumerate(plan[4:]):
+            st = g.neighbor2(*sdiag)
+            rf = g.neighbor2(*rdiag)
+            if st is not None or rf is not None:
+                cor_ops.append((d, st, rf))
                 if rf is not None:
-                    ext.recv(rbuf, rf, comm_id)
-            ext.group_end()
-            for st, rf, _, rbuf, ridx in pend:
-                if rf is not None:
-                    ext.unpack_cols(fields, rbuf, ridx)
+                    cor_mask |= 1 << d
+        if cor_ops:
+            ext.pack_corners(fb["cor_sbuf"][:4 * nf], fields)
 
-        # rows half: full rows are contiguous — RCCL moves them in place
-        # (zero-copy), both directions and all fields in ONE group
-        rows = [(g.neighbor(sdir), g.neighbor(rdir), ridx, sidx)
-                for _, sdir, rdir, ridx, sidx in sched[2:]]
-        if any(st is not None or rf is not None for st, rf, _, _ in rows):
-            comm_id = self.comm.rccl_handle()
-            ext.group_start()
-            for st, rf, ridx, sidx in rows:
-                for f in fields:
-                    if st is not None:
-                        ext.send(f[sidx], st, comm_id)
-                    if rf is not None:
-                        ext.recv(f[ridx], rf, comm_id)
-            ext.group_end()
+        if not (col_ops or row_ops or cor_ops):
+            return
+        comm_id = self.comm.rccl_handle()
+        ext.group_start()
+        for st, rf, sbuf, rbuf, _ in col_ops:
+            if st is not None:
+                ext.send(sbuf, st, comm_id)
+            if rf is not None:
+                ext.recv(rbuf, rf, comm_id)
+        for st, rf, ridx, sidx in row_ops:
+            for f in fields:
+                if st is not None:
+                    ext.send(f[sidx, 1:nx - 1], st, comm_id)
+                if rf is not None:
+                    ext.recv(f[ridx, 1:nx - 1], rf, comm_id)
+        for d, st, rf in cor_ops:
+            if st is not None:
+                ext.send(fb["cor_sbuf"][d * nf:(d + 1) * nf], st, comm_id)
+            if rf is not None:
+                ext.recv(fb["cor_rbuf"][d * nf:(d + 1) * nf], rf, comm_id)
+        ext.group_end()
+
+        for st, rf, _, rbuf, ridx in col_ops:
+            if rf is not None:
+                ext.unpack_cols(fields, rbuf, ridx)
+        if cor_mask:
+            ext.unpack_corners(fields, fb["cor_rbuf"][:4 * nf], cor_mask)
 
     # ------------------------------------------------------------------
     def make_stepper(self, state, steps_per_call=2, use_graph=None):

@@ -26,6 +26,38 @@ def default_dims(size):
     return nproc_y, size // nproc_y
 
 
+def halo_plan(nx, ny):
+    """The halo-exchange transfer plan for a (ny, nx) array.
+
+    The reference demo exchanges full edges in clockwise W,N,E,S order so
+    later phases carry corner values (shallow_water.py:180-208).  This plan
+    is bit-identical but fully order-independent: full columns, *interior*
+    rows, and the four halo corners exchanged directly with the diagonal
+    neighbors (a corner equals the diagonal neighbor's interior corner
+    cell).  Order-independence is what lets the fused GPU path issue the
+    entire exchange as ONE RCCL group.
+
+    Entries:
+      ("cols",   send_dir, recv_dir, recv_col, send_col)
+      ("rows",   send_dir, recv_dir, recv_row, send_row)   # cols 1..nx-2
+      ("corner", send_diag, recv_diag, recv_cell, send_cell)
+    The only ordering requirement: column writes land before corner writes
+    (both touch the 4 corner cells; corners win) — the executors below and
+    in the fused path preserve that.
+    """
+    return (
+        ("cols", "west", "east", nx - 1, 1),
+        ("cols", "east", "west", 0, nx - 2),
+        ("rows", "north", "south", 0, ny - 2),
+        ("rows", "south", "north", ny - 1, 1),
+        # recv corner (rj, ri) <- opposite diagonal's interior corner
+        ("corner", (-1, -1), (1, 1), (ny - 1, nx - 1), (1, 1)),
+        ("corner", (-1, 1), (1, -1), (ny - 1, 0), (1, nx - 2)),
+        ("corner", (1, -1), (-1, 1), (0, nx - 1), (ny - 2, 1)),
+        ("corner", (1, 1), (-1, -1), (0, 0), (ny - 2, nx - 2)),
+    )
+
+
 class CartesianGrid:
     """A 2-D process grid over a communicator.
 
@@ -51,23 +83,22 @@ class CartesianGrid:
     def rank_at(self, iy, ix):
         return iy * self.nproc_x + ix
 
+    _DIRS = {"south": (-1, 0), "north": (1, 0), "west": (0, -1),
+             "east": (0, 1)}
+
     def neighbor(self, direction):
         """Rank of the neighbor in a direction, or None at a closed edge.
 
         Directions follow the reference demo's convention
         (shallow_water.py:210-224): y grows northward, x grows eastward.
         """
-        iy, ix = self.coords
-        if direction == "south":
-            iy2, ix2 = iy - 1, ix
-        elif direction == "north":
-            iy2, ix2 = iy + 1, ix
-        elif direction == "west":
-            iy2, ix2 = iy, ix - 1
-        elif direction == "east":
-            iy2, ix2 = iy, ix + 1
-        else:
-            raise ValueError(f"bad direction {direction}")
+        dy, dx = self._DIRS[direction]
+        return self.neighbor2(dy, dx)
+
+    def neighbor2(self, dy, dx):
+        """Rank at a (possibly diagonal) grid offset, or None when closed."""
+        iy2 = self.coords[0] + dy
+        ix2 = self.coords[1] + dx
         if not 0 <= iy2 < self.nproc_y:
             if not self.periodic_y:
                 return None
@@ -77,33 +108,6 @@ class CartesianGrid:
                 return None
             ix2 %= self.nproc_x
         return self.rank_at(iy2, ix2)
-
-    # halo slice tables for a (ny, nx) array with a 1-cell halo ring.
-    # The reference demo exchanges in clockwise W,N,E,S order
-    # (shallow_water.py:180-208); we use both column exchanges first, then
-    # both row exchanges — full-row sends then carry fresh column-halo
-    # corners, which satisfies every corner dependency the clockwise order
-    # satisfies (and makes the SW corner fresh too), while letting the
-    # fused GPU path batch each half into a single RCCL group.
-    _SEND_ROW = {"south": 1, "north": -2}
-    _RECV_ROW = {"south": 0, "north": -1}
-    _SEND_COL = {"west": 1, "east": -2}
-    _RECV_COL = {"west": 0, "east": -1}
-    _ORDER = (("west", "east"), ("east", "west"),
-              ("north", "south"), ("south", "north"))
-
-    def _get_edge(self, arr, direction, kind):
-        if direction in ("south", "north"):
-            row = (self._SEND_ROW if kind == "send" else self._RECV_ROW)[direction]
-            return arr[row, :]
-        col = (self._SEND_COL if kind == "send" else self._RECV_COL)[direction]
-        return arr[:, col]
-
-    def _set_edge(self, arr, direction, value):
-        if direction in ("south", "north"):
-            arr[self._RECV_ROW[direction], :] = value
-        else:
-            arr[:, self._RECV_COL[direction]] = value
 
     def halo_exchange(self, arr):
         """Exchange the 1-cell halo ring of a local 2-D array.
@@ -124,30 +128,47 @@ class CartesianGrid:
         from ..ops.recv import recv
 
         me = self.comm.rank
-        for send_dir, recv_dir in self._ORDER:
-            send_to = self.neighbor(send_dir)
-            recv_from = self.neighbor(recv_dir)
+        ny, nx = out.shape
+
+        def xchg(send_to, recv_from, send_view, recv_setter, template):
             if send_to is None and recv_from is None:
-                continue
-            if send_to == me and recv_from == me:
-                # periodic self-wrap: a plain on-device copy
-                self._set_edge(out, recv_dir,
-                               self._get_edge(out, send_dir, "send"))
-                continue
+                return
             if send_to is None:
-                got = recv(self._get_edge(out, recv_dir, "recv"),
-                           source=recv_from, comm=self.comm)
-                self._set_edge(out, recv_dir, got)
+                recv_setter(recv(template, source=recv_from, comm=self.comm))
             elif recv_from is None:
-                send(self._get_edge(out, send_dir, "send"), dest=send_to,
-                     comm=self.comm)
+                send(send_view, dest=send_to, comm=self.comm)
             else:
-                got = sendrecv(
-                    self._get_edge(out, send_dir, "send"),
-                    self._get_edge(out, recv_dir, "recv"),
-                    source=recv_from,
-                    dest=send_to,
-                    comm=self.comm,
-                )
-                self._set_edge(out, recv_dir, got)
+                recv_setter(sendrecv(send_view, template, source=recv_from,
+                                     dest=send_to, comm=self.comm))
+
+        for kind, *spec in halo_plan(nx, ny):
+            if kind == "cols":
+                sdir, rdir, ridx, sidx = spec
+                st, rf = self.neighbor(sdir), self.neighbor(rdir)
+                if st == me and rf == me:
+                    out[:, ridx] = out[:, sidx]  # periodic self-wrap copy
+                    continue
+
+                def set_col(v, ridx=ridx):
+                    out[:, ridx] = v
+
+                xchg(st, rf, out[:, sidx], set_col, out[:, ridx])
+            elif kind == "rows":
+                sdir, rdir, ridx, sidx = spec
+                st, rf = self.neighbor(sdir), self.neighbor(rdir)
+
+                def set_row(v, ridx=ridx):
+                    out[ridx, 1:-1] = v
+
+                xchg(st, rf, out[sidx, 1:-1].contiguous(), set_row,
+                     out[ridx, 1:-1])
+            else:  # corner
+                (sdy, sdx), (rdy, rdx), (rj, ri), (sj, si) = spec
+                st, rf = self.neighbor2(sdy, sdx), self.neighbor2(rdy, rdx)
+
+                def set_corner(v, rj=rj, ri=ri):
+                    out[rj, ri] = v[0]
+
+                xchg(st, rf, out[sj, si:si + 1].contiguous(), set_corner,
+                     out[rj, ri:ri + 1])
         return out

@@ -61,76 +61,66 @@ def test_decomposed_matches_single_2x2():
     run_multiproc(_decomposed_matches_single, 4)
 
 
-def _phase_schedule_worker(rank, ws):
-    """Validate the fused halo phase plan against the eager exchange.
+def _clockwise_oracle(g, arr):
+    """The reference demo's exchange (full edges, clockwise W,N,E,S with
+    progressive updates — shallow_water.py:180-208), used as the oracle
+    for the order-independent halo_plan implementation."""
+    me = g.comm.rank
+    out = arr.clone()
+    edges = {
+        "west": (lambda a: a[:, 1], lambda a, v: a.__setitem__(
+            (slice(None), 0), v)),
+        "east": (lambda a: a[:, -2], lambda a, v: a.__setitem__(
+            (slice(None), -1), v)),
+        "south": (lambda a: a[1, :], lambda a, v: a.__setitem__(
+            (0, slice(None)), v)),
+        "north": (lambda a: a[-2, :], lambda a, v: a.__setitem__(
+            (-1, slice(None)), v)),
+    }
+    for sdir, rdir in (("west", "east"), ("north", "south"),
+                       ("east", "west"), ("south", "north")):
+        st, rf = g.neighbor(sdir), g.neighbor(rdir)
+        if st is None and rf is None:
+            continue
+        get_s, _ = edges[sdir]
+        get_r, set_r = edges[rdir]
+        if st == me and rf == me:
+            set_r(out, get_s(out))
+        elif st is None:
+            set_r(out, m.recv(get_r(out), source=rf, comm=g.comm))
+        elif rf is None:
+            m.send(get_s(out).contiguous(), dest=st, comm=g.comm)
+        else:
+            set_r(out, m.sendrecv(get_s(out).contiguous(), get_r(out),
+                                  source=rf, dest=st, comm=g.comm))
+    return out
 
-    The fused GPU path (ShallowWater._exchange_fields) follows
-    halo_phase_schedule(); executing the same plan with CPU ops must
-    reproduce CartesianGrid.halo_exchange exactly, corners included.
-    """
+
+def _halo_plan_vs_oracle(rank, ws):
+    """The order-independent plan (cols + interior rows + diagonal
+    corners) must reproduce the reference's clockwise full-edge exchange
+    exactly — corners included — for every periodic-x topology."""
     from mpi4jax_amd.parallel.grid import CartesianGrid
-    from mpi4jax_amd.models.shallow_water import halo_phase_schedule
 
-    for dims, periodic_x in (((2, 2), True), ((2, 2), False),
-                             ((2, 1), True), ((1, 2), True)):
+    for dims in ((2, 2), (2, 1), (1, 2), (ws, 1), (1, ws)):
         if dims[0] * dims[1] != ws:
             continue
-        g = CartesianGrid(m.get_world(), dims=dims,
-                          periodic=(False, periodic_x))
+        g = CartesianGrid(m.get_world(), dims=dims, periodic=(False, True))
         torch.manual_seed(100 + rank)
-        ny, nx = 6, 7
-        f0 = torch.randn(ny, nx)
-        f1 = torch.randn(ny, nx)
-        expect = [g.halo_exchange(f0), g.halo_exchange(f1)]
-
-        fields = [f0.clone(), f1.clone()]
-        me = g.comm.rank
-        for cols, sdir, rdir, ridx, sidx in halo_phase_schedule(nx, ny):
-            st, rf = g.neighbor(sdir), g.neighbor(rdir)
-            if st is None and rf is None:
-                continue
-            if st == me and rf == me:
-                for f in fields:
-                    if cols:
-                        f[:, ridx] = f[:, sidx]
-                    else:
-                        f[ridx, :] = f[sidx, :]
-                continue
-            if cols:
-                sbuf = torch.cat([f[:, sidx] for f in fields])
-                tmpl = torch.empty(len(fields) * ny)
-                if st is not None and rf is not None:
-                    rbuf = m.sendrecv(sbuf, tmpl, source=rf, dest=st,
-                                      comm=g.comm)
-                elif st is not None:
-                    m.send(sbuf, st, comm=g.comm)
-                    rbuf = None
-                else:
-                    rbuf = m.recv(tmpl, rf, comm=g.comm)
-                if rf is not None:
-                    for i, f in enumerate(fields):
-                        f[:, ridx] = rbuf[i * ny:(i + 1) * ny]
-            else:
-                for f in fields:
-                    if st is not None and rf is not None:
-                        got = m.sendrecv(f[sidx, :], f[ridx, :], source=rf,
-                                         dest=st, comm=g.comm)
-                        f[ridx, :] = got
-                    elif st is not None:
-                        m.send(f[sidx, :].contiguous(), st, comm=g.comm)
-                    else:
-                        f[ridx, :] = m.recv(f[ridx, :], rf, comm=g.comm)
-        for got, exp in zip(fields, expect):
-            assert torch.equal(got, exp), (dims, periodic_x, rank,
+        for ny, nx in ((6, 7), (5, 5)):
+            a = torch.randn(ny, nx)
+            got = g.halo_exchange(a)
+            exp = _clockwise_oracle(g, a)
+            assert torch.equal(got, exp), (dims, rank,
                                            (got - exp).abs().max())
 
 
-def test_fused_halo_phase_schedule_2x2():
-    run_multiproc(_phase_schedule_worker, 4)
+def test_halo_plan_vs_clockwise_oracle_4ranks():
+    run_multiproc(_halo_plan_vs_oracle, 4)
 
 
-def test_fused_halo_phase_schedule_2ranks():
-    run_multiproc(_phase_schedule_worker, 2)
+def test_halo_plan_vs_clockwise_oracle_2ranks():
+    run_multiproc(_halo_plan_vs_oracle, 2)
 
 
 def _two_rank(rank, ws):
