@@ -306,8 +306,13 @@ class ShallowWater:
 
         # resolve peers; pre-group kernels (col/corner packs, wraps)
         col_ops, row_ops, cor_ops = [], [], []
+        col_nbrs = [(g.neighbor(sdir), g.neighbor(rdir))
+                    for _, sdir, rdir, _, _ in plan[:2]]
+        if all(st == me and rf == me for st, rf in col_nbrs):
+            ext.halo_wrap(fields, 2)  # both periodic wraps, one kernel
+            col_nbrs = [(None, None), (None, None)]
         for k, (_, sdir, rdir, ridx, sidx) in enumerate(plan[:2]):
-            st, rf = g.neighbor(sdir), g.neighbor(rdir)
+            st, rf = col_nbrs[k]
             if st is None and rf is None:
                 continue
             if st == me and rf == me:
