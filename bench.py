@@ -35,8 +35,15 @@ def measure_allreduce_gbps(steps=20, warmup=5):
     nbytes = 256 * 1024 * 1024
     n = nbytes // 2
     x = torch.randn(n, device="cuda").to(torch.bfloat16)
+    # register the persistent buffer so RCCL can use zero-copy protocols
+    try:
+        from mpi4jax_amd._backend import rccl
+
+        rccl.ext().comm_register(comm.rccl_handle(), x)
+    except Exception:
+        pass
     for _ in range(warmup):
-        x = m.allreduce(x, m.SUM)
+        m.allreduce(x, m.SUM)
     m.barrier()
     torch.cuda.synchronize()
     t0 = time.perf_counter()

@@ -209,6 +209,22 @@ void comm_abort(int64_t id) {
   g_comms.erase(it);
 }
 
+// user-buffer registration (RCCL >= 2.19): lets RCCL use zero-copy
+// protocols on registered persistent buffers (multi-GPU bandwidth win)
+int64_t comm_register(int64_t comm_id, at::Tensor buf) {
+  auto c = get_comm(comm_id);
+  TORCH_CHECK(buf.is_cuda() && buf.is_contiguous(), "bad buffer");
+  void* handle = nullptr;
+  RCCL_CHECK(ncclCommRegister(c.comm, buf.data_ptr(),
+                              buf.numel() * buf.element_size(), &handle));
+  return (int64_t)(uintptr_t)handle;
+}
+
+void comm_deregister(int64_t comm_id, int64_t handle) {
+  auto c = get_comm(comm_id);
+  RCCL_CHECK(ncclCommDeregister(c.comm, (void*)(uintptr_t)handle));
+}
+
 void set_logging(bool enabled) { g_logging = enabled; }
 
 py::dict version_info() {
@@ -585,6 +601,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("comm_count", &comm_count);
   m.def("check_async_errors", &check_async_errors);
   m.def("comm_abort", &comm_abort);
+  m.def("comm_register", &comm_register);
+  m.def("comm_deregister", &comm_deregister);
   m.def("set_logging", &set_logging);
   m.def("version_info", &version_info);
   m.def("allreduce", &allreduce);

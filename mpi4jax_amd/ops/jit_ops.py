@@ -206,3 +206,61 @@ torch.library.register_autograd("mpi4jax_amd::sendrecv", _sendrecv_bwd,
 
 def sendrecv(sendbuf, recvbuf, source, dest, *, comm=None):
     return _sendrecv(sendbuf, recvbuf, source, dest, comm_key(comm))
+
+
+# ----------------------------------------------------------------- reduce
+@torch.library.custom_op("mpi4jax_amd::reduce", mutates_args=())
+def _reduce(x: torch.Tensor, op: str, root: int, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    out = backend_for(x).reduce(x.contiguous(), _OPS[op], root, comm)
+    return out if out is not None else x.clone()
+
+
+@_reduce.register_fake
+def _(x, op, root, key):
+    return torch.empty_like(x)
+
+
+def reduce(x, op, root, *, comm=None):
+    op = op.value if isinstance(op, Op) else str(op)
+    return _reduce(x, op, root, comm_key(comm))
+
+
+# ----------------------------------------------------------------- gather
+@torch.library.custom_op("mpi4jax_amd::gather", mutates_args=())
+def _gather(x: torch.Tensor, root: int, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    out = backend_for(x).gather(x.contiguous(), root, comm)
+    if out is None:  # non-root: keep a static output shape for the graph
+        out = x.new_empty((comm.size,) + tuple(x.shape))
+    return out
+
+
+@_gather.register_fake
+def _(x, root, key):
+    return x.new_empty((_comm(key).size,) + tuple(x.shape))
+
+
+def gather(x, root, *, comm=None):
+    """Root gets ``(nproc, *shape)``; other ranks get an undefined tensor
+    of that shape (static shapes are required inside a compiled graph —
+    the eager op's non-root input passthrough does not translate)."""
+    return _gather(x, root, comm_key(comm))
+
+
+# ----------------------------------------------------------------- scatter
+@torch.library.custom_op("mpi4jax_amd::scatter", mutates_args=())
+def _scatter(x: torch.Tensor, root: int, key: int) -> torch.Tensor:
+    comm = _comm(key)
+    return backend_for(x).scatter(x.contiguous(), root, comm)
+
+
+@_scatter.register_fake
+def _(x, root, key):
+    comm = _comm(key)
+    shape = tuple(x.shape[1:]) if comm.rank == root else tuple(x.shape)
+    return x.new_empty(shape)
+
+
+def scatter(x, root, *, comm=None):
+    return _scatter(x, root, comm_key(comm))

@@ -79,3 +79,15 @@ def test_compiled_allreduce_gpu():
     y = f(x)
     torch.cuda.synchronize()
     assert torch.equal(y, x + 1)
+
+
+def test_reduce_gather_scatter_compiled():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        r = jit_ops.reduce(x, "sum", 0)
+        g = jit_ops.gather(x, 0)[0]
+        s = jit_ops.scatter(x[None], 0)
+        return r + g + s
+
+    x = torch.arange(5.0)
+    assert torch.equal(f(x), 3 * x)
