@@ -30,6 +30,7 @@ class _Sendrecv(torch.autograd.Function):
          backend) = inputs
         ctx.meta = (source, dest, sendtag, recvtag, comm, backend)
         ctx.send_shape = tuple(sendbuf.shape)
+        ctx.recv_shape = tuple(recvbuf.shape)
 
     @staticmethod
     def backward(ctx, grad):
@@ -51,7 +52,8 @@ class _Sendrecv(torch.autograd.Function):
             raise RuntimeError(
                 "sendrecv jvp requires a tangent for sendbuf"
             )
-        template = send_t.new_empty(ctx.send_shape)
+        # the tangent arrives shaped like the OUTPUT (= recvbuf)
+        template = send_t.new_empty(ctx.recv_shape)
         return backend.sendrecv(
             send_t.contiguous(), template, source, dest, sendtag, recvtag,
             comm, None,

@@ -60,3 +60,17 @@ def test_chained_ops_grad():
     y = m.allreduce(A @ x, m.SUM)
     y.sum().backward()
     assert torch.allclose(x.grad, A.t() @ torch.ones(3))
+
+
+def test_sendrecv_jvp_self():
+    """Forward-mode tangent flows along the same edge; shapes can differ
+    between send and recv buffers."""
+    x = torch.randn(2, 3)
+    t = torch.randn(2, 3)
+    tmpl = torch.empty(6)  # different recv shape (reshaped message)
+    with fwd_ad.dual_level():
+        xd = fwd_ad.make_dual(x, t)
+        yd = m.sendrecv(xd, tmpl, source=0, dest=0)
+        y, yt = fwd_ad.unpack_dual(yd)
+    assert torch.equal(y, x.reshape(6))
+    assert torch.equal(yt, t.reshape(6))
