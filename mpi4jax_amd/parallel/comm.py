@@ -161,10 +161,6 @@ class Communicator:
             return self._rccl_id
 
 
-def _is_multiproc():
-    return _env_int("WORLD_SIZE", 1) > 1 or dist.is_initialized()
-
-
 def init(device=None):
     """Initialize the world communicator.
 
