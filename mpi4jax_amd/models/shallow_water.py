@@ -23,7 +23,8 @@ from collections import namedtuple
 import torch
 
 from ..parallel.comm import resolve_comm
-from ..parallel.grid import CartesianGrid, halo_plan
+from ..parallel.grid import (CartesianGrid, halo_plan,
+                             halo_exchange_schedule)
 
 ModelState = namedtuple("ModelState", "h u v dh du dv")
 
@@ -302,39 +303,14 @@ class ShallowWater:
             fb["cor_sbuf"] = torch.empty(12, dtype=self.dtype,
                                          device=self.device)
             fb["cor_rbuf"] = torch.empty_like(fb["cor_sbuf"])
-        plan = halo_plan(nx, ny)
+        (wrap_sides, col_ops, row_ops, cor_ops,
+         cor_mask) = halo_exchange_schedule(g, nx, ny)
 
-        # resolve peers; pre-group kernels (col/corner packs, wraps)
-        col_ops, row_ops, cor_ops = [], [], []
-        col_nbrs = [(g.neighbor(sdir), g.neighbor(rdir))
-                    for _, sdir, rdir, _, _ in plan[:2]]
-        if all(st == me and rf == me for st, rf in col_nbrs):
-            ext.halo_wrap(fields, 2)  # both periodic wraps, one kernel
-            col_nbrs = [(None, None), (None, None)]
-        for k, (_, sdir, rdir, ridx, sidx) in enumerate(plan[:2]):
-            st, rf = col_nbrs[k]
-            if st is None and rf is None:
-                continue
-            if st == me and rf == me:
-                ext.halo_wrap(fields, 0 if ridx == nx - 1 else 1)
-                continue
-            sbuf = fb[f"col_sbuf{k}"][:nf * ny]
-            rbuf = fb[f"col_rbuf{k}"][:nf * ny]
+        for side in wrap_sides:
+            ext.halo_wrap(fields, side)
+        for k, st, rf, sidx, _ in col_ops:
             if st is not None:
-                ext.pack_cols(sbuf, fields, sidx)
-            col_ops.append((st, rf, sbuf, rbuf, ridx))
-        for (_, sdir, rdir, ridx, sidx) in plan[2:4]:
-            st, rf = g.neighbor(sdir), g.neighbor(rdir)
-            if st is not None or rf is not None:
-                row_ops.append((st, rf, ridx, sidx))
-        cor_mask = 0
-        for d, (_, sdiag, rdiag, _, _) in enumerate(plan[4:]):
-            st = g.neighbor2(*sdiag)
-            rf = g.neighbor2(*rdiag)
-            if st is not None or rf is not None:
-                cor_ops.append((d, st, rf))
-                if rf is not None:
-                    cor_mask |= 1 << d
+                ext.pack_cols(fb[f"col_sbuf{k}"][:nf * ny], fields, sidx)
         if cor_ops:
             ext.pack_corners(fb["cor_sbuf"][:4 * nf], fields)
 
@@ -342,11 +318,11 @@ class ShallowWater:
             return
         comm_id = self.comm.rccl_handle()
         ext.group_start()
-        for st, rf, sbuf, rbuf, _ in col_ops:
+        for k, st, rf, _, _ in col_ops:
             if st is not None:
-                ext.send(sbuf, st, comm_id)
+                ext.send(fb[f"col_sbuf{k}"][:nf * ny], st, comm_id)
             if rf is not None:
-                ext.recv(rbuf, rf, comm_id)
+                ext.recv(fb[f"col_rbuf{k}"][:nf * ny], rf, comm_id)
         for st, rf, ridx, sidx in row_ops:
             for f in fields:
                 if st is not None:
@@ -360,9 +336,9 @@ class ShallowWater:
                 ext.recv(fb["cor_rbuf"][d * nf:(d + 1) * nf], rf, comm_id)
         ext.group_end()
 
-        for st, rf, _, rbuf, ridx in col_ops:
+        for k, st, rf, _, ridx in col_ops:
             if rf is not None:
-                ext.unpack_cols(fields, rbuf, ridx)
+                ext.unpack_cols(fields, fb[f"col_rbuf{k}"][:nf * ny], ridx)
         if cor_mask:
             ext.unpack_corners(fields, fb["cor_rbuf"][:4 * nf], cor_mask)
 

@@ -58,6 +58,50 @@ def halo_plan(nx, ny):
     )
 
 
+def halo_exchange_schedule(grid, nx, ny):
+    """Resolve halo_plan against a grid into concrete transfer ops.
+
+    Returns ``(wrap_sides, col_ops, row_ops, cor_ops, cor_mask)``:
+      wrap_sides: kernel-side periodic wraps (0=east halo, 1=west, 2=both)
+      col_ops:  [(k, send_to, recv_from, send_col, recv_col)]
+      row_ops:  [(send_to, recv_from, recv_row, send_row)]  (interior cols)
+      cor_ops:  [(d, send_to, recv_from)]  (diagonal corner transfers)
+    All peer ranks are comm-relative; every transfer list entry has at
+    least one non-None peer.  The cross-rank matching of this schedule
+    (send sequences == recv sequences per ordered rank pair) is verified
+    for every topology in tests/test_shallow_water.py.
+    """
+    me = grid.comm.rank
+    plan = halo_plan(nx, ny)
+    wrap_sides, col_ops, row_ops, cor_ops = [], [], [], []
+    col_nbrs = [(grid.neighbor(sdir), grid.neighbor(rdir))
+                for _, sdir, rdir, _, _ in plan[:2]]
+    if all(st == me and rf == me for st, rf in col_nbrs):
+        wrap_sides.append(2)  # both periodic wraps in one kernel
+        col_nbrs = [(None, None), (None, None)]
+    for k, (_, sdir, rdir, ridx, sidx) in enumerate(plan[:2]):
+        st, rf = col_nbrs[k]
+        if st is None and rf is None:
+            continue
+        if st == me and rf == me:
+            wrap_sides.append(0 if ridx == nx - 1 else 1)
+            continue
+        col_ops.append((k, st, rf, sidx, ridx))
+    for (_, sdir, rdir, ridx, sidx) in plan[2:4]:
+        st, rf = grid.neighbor(sdir), grid.neighbor(rdir)
+        if st is not None or rf is not None:
+            row_ops.append((st, rf, ridx, sidx))
+    cor_mask = 0
+    for d, (_, sdiag, rdiag, _, _) in enumerate(plan[4:]):
+        st = grid.neighbor2(*sdiag)
+        rf = grid.neighbor2(*rdiag)
+        if st is not None or rf is not None:
+            cor_ops.append((d, st, rf))
+            if rf is not None:
+                cor_mask |= 1 << d
+    return wrap_sides, col_ops, row_ops, cor_ops, cor_mask
+
+
 class CartesianGrid:
     """A 2-D process grid over a communicator.
 
