@@ -16,10 +16,17 @@ import torch
 
 
 def barrier(*, comm=None, token=NOTSET):
-    """Synchronize all processes of ``comm``."""
+    """Synchronize all processes of ``comm``.
+
+    Backend choice must be identical on every rank: the RCCL (stream)
+    barrier is used only when this communicator already has an RCCL
+    communicator — i.e. the program has been doing GPU collectives on it.
+    Otherwise the bootstrap-plane (host) barrier runs; that also keeps
+    CPU-only multi-process runs correct on machines that have a GPU.
+    """
     raise_if_token_is_set(token)
     comm = resolve_comm(comm)
-    if torch.cuda.is_available() and torch.cuda.is_initialized():
+    if comm._rccl_id is not None and torch.cuda.is_initialized():
         from .._backend import rccl
 
         rccl.barrier(comm)
