@@ -138,55 +138,7 @@ __global__ void sw_stage1_kernel(SwArgs<T> a) {
   }
 }
 
-// stage 2: tendencies dnh/dnu/dnv on the interior
-template <typename T>
-__global__ void sw_stage2_kernel(SwArgs<T> a) {
-  const int ny = (int)a.ny, nx = (int)a.nx;
-  SW_BLOCK_MAP(ny, nx);
-  if (i < nx && j < ny) {
-    const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
-
-    T dnh = -(a.fe[idx] - a.fe[idx - 1]) / a.dx -
-            (a.fn[idx] - a.fn[idx - nx]) / a.dy;
-    a.dnh[idx] = dnh;
-
-    T dnu = -G * (a.h[idx + 1] - a.h[idx]) / a.dx +
-            T(0.5) * (a.q[idx] * T(0.5) * (a.fn[idx] + a.fn[idx + 1]) +
-                      a.q[idx - nx] * T(0.5) *
-                          (a.fn[idx - nx] + a.fn[idx - nx + 1]));
-    dnu -= (a.ke[idx + 1] - a.ke[idx]) / a.dx;
-    a.dnu[idx] = dnu;
-
-    T dnv = -G * (a.h[idx + nx] - a.h[idx]) / a.dy -
-            T(0.5) * (a.q[idx] * T(0.5) * (a.fe[idx] + a.fe[idx + nx]) +
-                      a.q[idx - 1] * T(0.5) *
-                          (a.fe[idx - 1] + a.fe[idx + nx - 1]));
-    dnv -= (a.ke[idx + nx] - a.ke[idx]) / a.dy;
-    a.dnv[idx] = dnv;
-  }
-}
-
-// stage 3: Adams-Bashforth (or Euler) time update, own-cell, in place
-template <typename T>
-__global__ void sw_stage3_kernel(SwArgs<T> a) {
-  const int ny = (int)a.ny, nx = (int)a.nx;
-  SW_BLOCK_MAP(ny, nx);
-  if (i < nx && j < ny) {
-    const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
-    T uu = a.u[idx] + a.dt * (a.ab_a * a.dnu[idx] + a.ab_b * a.dou[idx]);
-    T vv = a.v[idx] + a.dt * (a.ab_a * a.dnv[idx] + a.ab_b * a.dov[idx]);
-    a.h[idx] += a.dt * (a.ab_a * a.dnh[idx] + a.ab_b * a.doh[idx]);
-    if (a.f.east_wall && i == nx - 2) uu = T(0);
-    if (a.f.north_wall && j == ny - 2) vv = T(0);
-    a.u[idx] = uu;
-    a.v[idx] = vv;
-  }
-}
-
-// stage 4: lateral-friction Laplacians lu/lv (reusing fe/fn as scratch
-// would race; dedicated lu/lv buffers are passed via fe/fn slots)
+// lateral-friction gradient helpers (masked per the eager eb() semantics)
 template <typename T>
 __device__ inline T gu_of_u(const SwArgs<T>& a, int j, int i) {
   // gu = nu * du/dx, u-kind halos: west col open-only, others unread
@@ -223,41 +175,6 @@ __device__ inline T gv_of_v(const SwArgs<T>& a, int j, int i) {
   if (a.f.north_wall && j == ny - 2) return T(0);
   return a.nu * (a.v[(j + 1) * nx + i] - a.v[j * nx + i]) / a.dy;
 }
-
-template <typename T>
-__global__ void sw_stage4_kernel(SwArgs<T> a) {
-  const int ny = (int)a.ny, nx = (int)a.nx;
-  T* lu = a.fe;  // scratch reuse
-  T* lv = a.fn;
-  SW_BLOCK_MAP(ny, nx);
-  if (i < nx && j < ny) {
-    const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
-    lu[idx] = (gu_of_u(a, j, i) - gu_of_u(a, j, i - 1)) / a.dx +
-              (gv_of_u(a, j, i) - gv_of_u(a, j - 1, i)) / a.dy;
-    lv[idx] = (gu_of_v(a, j, i) - gu_of_v(a, j, i - 1)) / a.dx +
-              (gv_of_v(a, j, i) - gv_of_v(a, j - 1, i)) / a.dy;
-  }
-}
-
-template <typename T>
-__global__ void sw_stage5_kernel(SwArgs<T> a) {
-  const int ny = (int)a.ny, nx = (int)a.nx;
-  const T* lu = a.fe;
-  const T* lv = a.fn;
-  SW_BLOCK_MAP(ny, nx);
-  if (i < nx && j < ny) {
-    const int idx = j * nx + i;
-    if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) return;
-    T uu = a.u[idx] + a.dt * lu[idx];
-    T vv = a.v[idx] + a.dt * lv[idx];
-    if (a.f.east_wall && i == nx - 2) uu = T(0);
-    if (a.f.north_wall && j == ny - 2) vv = T(0);
-    a.u[idx] = uu;
-    a.v[idx] = vv;
-  }
-}
-
 
 // masked derived-field evaluators: single source of truth for the fe/fn/
 // q/ke formulas AND their open/closed halo masks (same semantics as the
@@ -367,7 +284,7 @@ __global__ void sw_stage8_kernel(SwArgs<T> a) {
   }
 }
 
-// stage 6 = stage2 + stage3 merged with double-buffered field output:
+// stage 6 = tendencies + AB time update merged with double-buffered field output:
 // tendencies + AB/Euler update written to h2/u2/v2 — no in-place hazard,
 // one full pass of traffic saved.
 template <typename T>
@@ -868,10 +785,6 @@ static void sw_launch(int stage, const SwLaunchParams& p,
   dim3 grid((unsigned)(gx * p.ny)), block(kBlock);
   switch (stage) {
     case 1: hipLaunchKernelGGL(sw_stage1_kernel<T>, grid, block, 0, stream, a); break;
-    case 2: hipLaunchKernelGGL(sw_stage2_kernel<T>, grid, block, 0, stream, a); break;
-    case 3: hipLaunchKernelGGL(sw_stage3_kernel<T>, grid, block, 0, stream, a); break;
-    case 4: hipLaunchKernelGGL(sw_stage4_kernel<T>, grid, block, 0, stream, a); break;
-    case 5: hipLaunchKernelGGL(sw_stage5_kernel<T>, grid, block, 0, stream, a); break;
     case 6: hipLaunchKernelGGL(sw_stage6_kernel<T>, grid, block, 0, stream, a); break;
     case 7: hipLaunchKernelGGL(sw_stage7_kernel<T>, grid, block, 0, stream, a); break;
     case 8: hipLaunchKernelGGL(sw_stage8_kernel<T>, grid, block, 0, stream, a); break;
