@@ -12,8 +12,11 @@ The reference's benchmark config (docs/shallow-water.rst:49-52) is a
 (3600, 1800) float32 domain run for 0.1 model days; wall-clock per model
 day is the headline metric.
 
-Implementation is eager torch ops; the ``fused`` path (HIP kernels +
-hipGraph capture) plugs in at :meth:`step` without changing semantics.
+Two step implementations share semantics: an eager torch-op path (CPU
+and reference/debugging) and the fused MI355X path — two or three CDNA4
+kernels (stage 18: tendencies+update; stage 17: friction; optional scalar
+fallbacks) plus a single-group halo exchange, with hipGraph capture for
+multistep replay.
 """
 
 import math
@@ -102,8 +105,8 @@ class ShallowWater:
         # CFL time step (shallow_water.py:135)
         self.dt = 0.125 * min(self.dx, self.dy) / math.sqrt(GRAVITY * DEPTH)
 
-        # fused CDNA4 kernel path (GPU only): 5 stencil kernels + h/u/v
-        # halo exchange per step instead of ~300 eager torch kernels
+        # fused CDNA4 kernel path (GPU only): 2-3 stencil kernels + one
+        # halo-exchange group per step instead of ~300 eager torch kernels
         if fused is None:
             fused = self.device.type == "cuda"
         self.fused = bool(fused) and self.device.type == "cuda"
@@ -181,7 +184,7 @@ class ShallowWater:
         return self._step_eager(state, first_step)
 
     # ------------------------------------------------------------------
-    # fused GPU path: csrc/shallow_water.hip (5 stencil kernels) + in-place
+    # fused GPU path: csrc/shallow_water.hip stencil kernels + in-place
     # halo exchange of h/u/v.  Produces the same solution as the eager path
     # (equal up to FMA contraction inside the fused kernels).
     def _fused_flags(self):
@@ -276,10 +279,6 @@ class ShallowWater:
                           fb["do_v"])
 
     # ------------------------------------------------------------------
-    # fused halo exchange: one CDNA4 kernel per phase (wrap / column
-    # pack+unpack) + grouped RCCL p2p; row phases send/recv directly from
-    # field memory (rows are contiguous — true zero-copy).  Phase order
-    # matches the eager path (W,N,E,S) so halo corners compose identically.
     def _exchange_fields(self, fields):
         """One-group halo exchange (halo_plan is order-independent).
 
