@@ -9,3 +9,9 @@ from .comm import (  # noqa: F401
 )
 from .grid import CartesianGrid  # noqa: F401
 from .ddp import average_gradients  # noqa: F401
+from .tp import (  # noqa: F401
+    ColumnParallelLinear,
+    RowParallelLinear,
+    copy_to_parallel,
+)
+from .sp import seq_to_head_shard, head_to_seq_shard  # noqa: F401
