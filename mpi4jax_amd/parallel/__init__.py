@@ -15,3 +15,8 @@ from .tp import (  # noqa: F401
     copy_to_parallel,
 )
 from .sp import seq_to_head_shard, head_to_seq_shard  # noqa: F401
+from .pp import (  # noqa: F401
+    send_activation,
+    recv_activation,
+    backward_send,
+)
