@@ -334,3 +334,22 @@ def test_allreduce_bandwidth_smoke():
     y = m.allreduce(xx, m.SUM)
     torch.cuda.synchronize()
     assert torch.equal(y[:1000].cpu(), xx[:1000].cpu())
+
+
+def test_parallel_helpers_gpu_single_rank():
+    """TP layers and SP shard swaps run on the GPU op path at world 1."""
+    from mpi4jax_amd.parallel import (ColumnParallelLinear,
+                                      RowParallelLinear, seq_to_head_shard,
+                                      head_to_seq_shard, average_gradients)
+
+    col = ColumnParallelLinear(8, 6).cuda()
+    row = RowParallelLinear(6, 4).cuda()
+    x = torch.randn(3, 8, device="cuda", requires_grad=True)
+    y = row(torch.tanh(col(x)))
+    y.sum().backward()
+    torch.cuda.synchronize()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    average_gradients([p for p in col.parameters()])
+
+    t = torch.randn(4, 2, 3, device="cuda")
+    assert torch.equal(head_to_seq_shard(seq_to_head_shard(t)), t)
