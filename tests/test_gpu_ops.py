@@ -353,3 +353,15 @@ def test_parallel_helpers_gpu_single_rank():
 
     t = torch.randn(4, 2, 3, device="cuda")
     assert torch.equal(head_to_seq_shard(seq_to_head_shard(t)), t)
+
+
+def test_huge_allreduce_64bit_counts():
+    """> 2^31 elements: exercises 64-bit counts end to end (8 GiB bf16)."""
+    n = (1 << 32) + 4096  # 4.29e9 elements = 8.6 GB
+    x = torch.ones(n, dtype=torch.bfloat16, device="cuda")
+    y = m.allreduce(x, m.SUM)
+    torch.cuda.synchronize()
+    assert y.numel() == n
+    assert y[0].item() == 1.0 and y[-1].item() == 1.0
+    del x, y
+    torch.cuda.empty_cache()
