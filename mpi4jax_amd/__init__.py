@@ -50,6 +50,7 @@ from ._backend import (  # noqa: F401
     has_rccl_support,
     has_cuda_support,
     has_sycl_support,
+    version_info,
 )
 from .utils.logging import set_logging, get_logging  # noqa: F401
 
@@ -87,6 +88,7 @@ __all__ = [
     "has_rccl_support",
     "has_cuda_support",
     "has_sycl_support",
+    "version_info",
     "Op",
     "SUM",
     "PROD",

@@ -46,3 +46,14 @@ def has_sycl_support() -> bool:
     """Always False — single-vendor by design (drop-in parity with the
     reference's capability probe, utils.py:159-174)."""
     return False
+
+
+def version_info() -> dict:
+    """RCCL / HIP runtime versions (diagnostic parity with the
+    reference's MPI_ABI_INFO export, mpi_xla_bridge_cpu.cpp:524-533)."""
+    try:
+        from . import rccl
+
+        return dict(rccl.ext().version_info())
+    except Exception:
+        return {}

@@ -18,6 +18,11 @@ Usage::
 Communicators cross the graph boundary as int registry keys (the same
 int64-handle marshalling idea the reference uses for MPI objects,
 ``mpi_ops_common.h:36-48``).
+
+Not traced: ``send``/``recv`` (their buffered self-messaging queue is
+process state a traced graph must not capture) and ``barrier`` (returns
+nothing; call it outside the compiled region).  Use :func:`sendrecv` for
+paired transfers inside graphs.
 """
 
 import torch
