@@ -70,6 +70,8 @@ struct SwArgs {
   T* v2;
   long long ny, nx;
   T dx, dy, dt, nu;
+  T rdx, rdy;  // hoisted reciprocals (vector kernels: f32 division is
+               // ~16-30 unpipelined cycles; uniform denominators multiply)
   T cor_base, cor_dj;  // coriolis(j) = cor_base + j * cor_dj
   T ab_a, ab_b;        // Adams-Bashforth coefficients (b=0 -> Euler)
   SwFlags f;
@@ -505,7 +507,7 @@ __global__ void sw_stage1v(SwArgs<float> a) {
   st4(a.fe, idx, 0.5f * (hA + hAe) * uc);
   st4(a.fn, idx, 0.5f * (hA + hB) * vc);
   float cor = a.cor_base + (float)j * a.cor_dj;
-  vf4 q = cor + ((ve - vc) / a.dx - (un - uc) / a.dy);
+  vf4 q = cor + ((ve - vc) * a.rdx - (un - uc) * a.rdy);
   q *= 1.f / (0.25f * (hA + hAe + hB + hBe));
   st4(a.q, idx, q);
   vf4 ke = 0.5f * (0.5f * (uc * uc + uw * uw) + 0.5f * (vc * vc + vs * vs));
@@ -547,15 +549,15 @@ __global__ void sw_stage6v(SwArgs<float> a) {
   vf4 uc = ld4(a.u, idx), vc = ld4(a.v, idx);
   vf4 doh = ld4(a.doh, idx), dou = ld4(a.dou, idx), dov = ld4(a.dov, idx);
 
-  vf4 dnh = -(fec - few) / a.dx - (fnc - fns) / a.dy;
+  vf4 dnh = -(fec - few) * a.rdx - (fnc - fns) * a.rdy;
   st4(a.dnh, idx, dnh);
-  vf4 dnu = -G * (he - hc) / a.dx +
+  vf4 dnu = -G * (he - hc) * a.rdx +
             0.5f * (qc * 0.5f * (fnc + fne) + qs * 0.5f * (fns + fnse));
-  dnu -= (kee - kec) / a.dx;
+  dnu -= (kee - kec) * a.rdx;
   st4(a.dnu, idx, dnu);
-  vf4 dnv = -G * (hn - hc) / a.dy -
+  vf4 dnv = -G * (hn - hc) * a.rdy -
             0.5f * (qc * 0.5f * (fec + fen) + qw * 0.5f * (few + fenw));
-  dnv -= (ken - kec) / a.dy;
+  dnv -= (ken - kec) * a.rdy;
   st4(a.dnv, idx, dnv);
 
   st4(a.h2, idx, hc + a.dt * (a.ab_a * dnh + a.ab_b * doh));
@@ -595,10 +597,10 @@ __global__ void sw_stage7v(SwArgs<float> a) {
   vf4 vcc = ld4(a.v, idx), vw = ld4(a.v, idx - 1), ve = ld4(a.v, idx + 1);
   vf4 vn = ld4(a.v, idx + nx), vs = ld4(a.v, idx - nx);
   const float nu = a.nu;
-  vf4 lu = (nu * (ue - uc) / a.dx - nu * (uc - uw) / a.dx) / a.dx +
-           (nu * (un - uc) / a.dy - nu * (uc - us) / a.dy) / a.dy;
-  vf4 lv = (nu * (ve - vcc) / a.dx - nu * (vcc - vw) / a.dx) / a.dx +
-           (nu * (vn - vcc) / a.dy - nu * (vcc - vs) / a.dy) / a.dy;
+  vf4 lu = (nu * (ue - uc) * a.rdx - nu * (uc - uw) * a.rdx) * a.rdx +
+           (nu * (un - uc) * a.rdy - nu * (uc - us) * a.rdy) * a.rdy;
+  vf4 lv = (nu * (ve - vcc) * a.rdx - nu * (vcc - vw) * a.rdx) * a.rdx +
+           (nu * (vn - vcc) * a.rdy - nu * (vcc - vs) * a.rdy) * a.rdy;
   st4(a.u2, idx, uc + a.dt * lu);
   st4(a.v2, idx, vcc + a.dt * lv);
 }
@@ -685,7 +687,7 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   vf4 V0m1 = V0A, V00 = sh0(V0A, V0B.x), V01 = sh1(V0A, V0B.x, V0B.y);
   vf4 Vp0 = ld4(a.v, idp);
 
-  const float dx = a.dx, dy = a.dy;
+  const float rdx = a.rdx, rdy = a.rdy;
 
   // derived fields (same formulas as fe_at/fn_at/q_at/ke_at, full-mask
   // region so every value is the plain formula)
@@ -700,11 +702,11 @@ __global__ void sw_stage18v(SwArgs<float> a) {
 
   float corj = a.cor_base + (float)j * a.cor_dj;
   float corjm = a.cor_base + (float)(j - 1) * a.cor_dj;
-  vf4 q_c = corj + ((V01 - V00) / dx - (Up0 - U00) / dy);
+  vf4 q_c = corj + ((V01 - V00) * rdx - (Up0 - U00) * rdy);
   q_c *= 1.f / (0.25f * (H00 + H01 + Hp0 + Hp1));
-  vf4 q_s = corjm + ((Vm1 - Vm0) / dx - (U00 - Um0) / dy);
+  vf4 q_s = corjm + ((Vm1 - Vm0) * rdx - (U00 - Um0) * rdy);
   q_s *= 1.f / (0.25f * (Hm0 + Hm1 + H00 + H01));
-  vf4 q_w = corj + ((V00 - V0m1) / dx - (Upm1 - U0m1) / dy);
+  vf4 q_w = corj + ((V00 - V0m1) * rdx - (Upm1 - U0m1) * rdy);
   q_w *= 1.f / (0.25f * (H0m1 + H00 + Hpm1 + Hp0));
 
   vf4 ke_c = 0.5f * (0.5f * (U00 * U00 + U0m1 * U0m1) +
@@ -714,15 +716,15 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   vf4 ke_n = 0.5f * (0.5f * (Up0 * Up0 + Upm1 * Upm1) +
                      0.5f * (Vp0 * Vp0 + V00 * V00));
 
-  vf4 dnh = -(fe_c - fe_w) / dx - (fn_c - fn_s) / dy;
-  vf4 dnu = -G * (H01 - H00) / dx +
+  vf4 dnh = -(fe_c - fe_w) * rdx - (fn_c - fn_s) * rdy;
+  vf4 dnu = -G * (H01 - H00) * rdx +
             0.5f * (q_c * 0.5f * (fn_c + fn_e) +
                     q_s * 0.5f * (fn_s + fn_se));
-  dnu -= (ke_e - ke_c) / dx;
-  vf4 dnv = -G * (Hp0 - H00) / dy -
+  dnu -= (ke_e - ke_c) * rdx;
+  vf4 dnv = -G * (Hp0 - H00) * rdy -
             0.5f * (q_c * 0.5f * (fe_c + fe_n) +
                     q_w * 0.5f * (fe_w + fe_nw));
-  dnv -= (ke_n - ke_c) / dy;
+  dnv -= (ke_n - ke_c) * rdy;
 
   st4(a.dnh, idx, dnh);
   st4(a.dnu, idx, dnu);
@@ -766,6 +768,8 @@ static void sw_fill_args(SwArgs<T>& a, const SwLaunchParams& p) {
   a.dy = (T)p.dy;
   a.dt = (T)p.dt;
   a.nu = (T)p.nu;
+  a.rdx = T(1) / (T)p.dx;
+  a.rdy = T(1) / (T)p.dy;
   a.cor_base = (T)p.cor_base;
   a.cor_dj = (T)p.cor_dj;
   a.ab_a = (T)p.ab_a;
