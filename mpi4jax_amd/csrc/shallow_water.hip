@@ -508,7 +508,7 @@ __global__ void sw_stage1v(SwArgs<float> a) {
   st4(a.fn, idx, 0.5f * (hA + hB) * vc);
   float cor = a.cor_base + (float)j * a.cor_dj;
   vf4 q = cor + ((ve - vc) * a.rdx - (un - uc) * a.rdy);
-  q *= 1.f / (0.25f * (hA + hAe + hB + hBe));
+  q *= rcp4(0.25f * (hA + hAe + hB + hBe));
   st4(a.q, idx, q);
   vf4 ke = 0.5f * (0.5f * (uc * uc + uw * uw) + 0.5f * (vc * vc + vs * vs));
   st4(a.ke, idx, ke);
@@ -616,6 +616,14 @@ __device__ inline vf2 ld2(const float* p, long long off) {
 }
 
 // shifted-vector builders: lane c of the result holds value at i0+c+k
+
+__device__ inline vf4 rcp4(vf4 x) {
+  // v_rcp_f32 per lane: ~1 ulp, removes three correctly-rounded-division
+  // chains (~30 unpipelined cycles each) from the per-pack critical path
+  return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
+               __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)};
+}
+
 __device__ inline vf4 sh0(vf4 Am1, float x3) {
   // offset 0 from a load at idx-1 plus the i0+3 element
   return (vf4){Am1.y, Am1.z, Am1.w, x3};
@@ -703,11 +711,11 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   float corj = a.cor_base + (float)j * a.cor_dj;
   float corjm = a.cor_base + (float)(j - 1) * a.cor_dj;
   vf4 q_c = corj + ((V01 - V00) * rdx - (Up0 - U00) * rdy);
-  q_c *= 1.f / (0.25f * (H00 + H01 + Hp0 + Hp1));
+  q_c *= rcp4(0.25f * (H00 + H01 + Hp0 + Hp1));
   vf4 q_s = corjm + ((Vm1 - Vm0) * rdx - (U00 - Um0) * rdy);
-  q_s *= 1.f / (0.25f * (Hm0 + Hm1 + H00 + H01));
+  q_s *= rcp4(0.25f * (Hm0 + Hm1 + H00 + H01));
   vf4 q_w = corj + ((V00 - V0m1) * rdx - (Upm1 - U0m1) * rdy);
-  q_w *= 1.f / (0.25f * (H0m1 + H00 + Hpm1 + Hp0));
+  q_w *= rcp4(0.25f * (H0m1 + H00 + Hpm1 + Hp0));
 
   vf4 ke_c = 0.5f * (0.5f * (U00 * U00 + U0m1 * U0m1) +
                      0.5f * (V00 * V00 + Vm0 * Vm0));
