@@ -475,6 +475,13 @@ __device__ inline void stage7_cell(const SwArgs<float>& a, int j, int i) {
 
 // simpler pack mapping: thread t covers row j = t / packs_per_row,
 // columns [4*(t % ppr), 4*(t % ppr)+3]
+
+__device__ inline vf4 rcp4(vf4 x) {
+  // v_rcp_f32 per lane: ~1 ulp, removes three correctly-rounded-division
+  // chains (~30 unpipelined cycles each) from the per-pack critical path
+  return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
+               __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)}
+
 __global__ void sw_stage1v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
@@ -617,11 +624,6 @@ __device__ inline vf2 ld2(const float* p, long long off) {
 
 // shifted-vector builders: lane c of the result holds value at i0+c+k
 
-__device__ inline vf4 rcp4(vf4 x) {
-  // v_rcp_f32 per lane: ~1 ulp, removes three correctly-rounded-division
-  // chains (~30 unpipelined cycles each) from the per-pack critical path
-  return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
-               __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)};
 }
 
 __device__ inline vf4 sh0(vf4 Am1, float x3) {
