@@ -380,6 +380,14 @@ __device__ inline void st4(float* p, long long off, vf4 v) {
   *(vf4*)(p + off) = v;
 }
 
+__device__ inline vf4 rcp4(vf4 x) {
+  // v_rcp_f32 per lane (~1 ulp): removes correctly-rounded-division
+  // chains (~30 unpipelined cycles each) from the per-pack critical path
+  return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
+               __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)};
+}
+
+
 // scalar per-cell bodies (shared by scalar kernels' fallback)
 __device__ inline void stage1_cell(const SwArgs<float>& a, int j, int i) {
   const int ny = (int)a.ny, nx = (int)a.nx;
@@ -475,13 +483,6 @@ __device__ inline void stage7_cell(const SwArgs<float>& a, int j, int i) {
 
 // simpler pack mapping: thread t covers row j = t / packs_per_row,
 // columns [4*(t % ppr), 4*(t % ppr)+3]
-
-__device__ inline vf4 rcp4(vf4 x) {
-  // v_rcp_f32 per lane: ~1 ulp, removes three correctly-rounded-division
-  // chains (~30 unpipelined cycles each) from the per-pack critical path
-  return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
-               __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)}
-
 __global__ void sw_stage1v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
@@ -623,9 +624,6 @@ __device__ inline vf2 ld2(const float* p, long long off) {
 }
 
 // shifted-vector builders: lane c of the result holds value at i0+c+k
-
-}
-
 __device__ inline vf4 sh0(vf4 Am1, float x3) {
   // offset 0 from a load at idx-1 plus the i0+3 element
   return (vf4){Am1.y, Am1.z, Am1.w, x3};
