@@ -639,7 +639,7 @@ __device__ inline vf4 shm1f(float xm1, vf4 A0) {
   return (vf4){xm1, A0.x, A0.y, A0.z};
 }
 
-__global__ __launch_bounds__(256, 5) void sw_stage18v(SwArgs<float> a) {
+__global__ void sw_stage18v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
   // XCD-aware block remap (same rationale as SW_BLOCK_MAP): give each XCD
