@@ -666,34 +666,36 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   const long long idx = (long long)j * nx + i0;
   const long long idn = idx - nx, idp = idx + nx;
 
-  // h rows j-1 (offsets 0,1), j and j+1 (offsets -1,0,1)
+  // issue ALL global loads first (one deep in-flight window, a single
+  // vmcnt wait region), then build the shifted vectors
   vf4 HmA = ld4(a.h, idn);
   float Hm4 = a.h[idn + 4];
   vf4 H0A = ld4(a.h, idx - 1);
   vf2 H0B = ld2(a.h, idx + 3);
   vf4 HpA = ld4(a.h, idp - 1);
   vf2 HpB = ld2(a.h, idp + 3);
-  vf4 Hm0 = sh0f(HmA), Hm1 = sh1f(HmA, Hm4);
-  vf4 H0m1 = H0A, H00 = sh0(H0A, H0B.x), H01 = sh1(H0A, H0B.x, H0B.y);
-  vf4 Hpm1 = HpA, Hp0 = sh0(HpA, HpB.x), Hp1 = sh1(HpA, HpB.x, HpB.y);
-
-  // u rows j-1 (offset 0), j (-1,0,1), j+1 (-1,0)
   vf4 Um0 = ld4(a.u, idn);
   vf4 U0A = ld4(a.u, idx - 1);
   vf2 U0B = ld2(a.u, idx + 3);
-  vf4 U0m1 = U0A, U00 = sh0(U0A, U0B.x), U01 = sh1(U0A, U0B.x, U0B.y);
   vf4 UpA = ld4(a.u, idp);
   float Upm1s = a.u[idp - 1];
-  vf4 Up0 = UpA, Upm1 = shm1f(Upm1s, UpA);
-
-  // v rows j-1 (0,1), j (-1,0,1), j+1 (0)
   vf4 VmA = ld4(a.v, idn);
   float Vm4 = a.v[idn + 4];
-  vf4 Vm0 = VmA, Vm1 = sh1f(VmA, Vm4);
   vf4 V0A = ld4(a.v, idx - 1);
   vf2 V0B = ld2(a.v, idx + 3);
-  vf4 V0m1 = V0A, V00 = sh0(V0A, V0B.x), V01 = sh1(V0A, V0B.x, V0B.y);
   vf4 Vp0 = ld4(a.v, idp);
+  vf4 doh = ld4(a.doh, idx), dou = ld4(a.dou, idx), dov = ld4(a.dov, idx);
+
+  // shifted-vector views: h rows j-1 (offsets 0,1), j / j+1 (-1,0,1);
+  // u rows j-1 (0), j (-1,0,1), j+1 (-1,0); v rows j-1 (0,1), j (-1,0,1),
+  // j+1 (0)
+  vf4 Hm0 = sh0f(HmA), Hm1 = sh1f(HmA, Hm4);
+  vf4 H0m1 = H0A, H00 = sh0(H0A, H0B.x), H01 = sh1(H0A, H0B.x, H0B.y);
+  vf4 Hpm1 = HpA, Hp0 = sh0(HpA, HpB.x), Hp1 = sh1(HpA, HpB.x, HpB.y);
+  vf4 U0m1 = U0A, U00 = sh0(U0A, U0B.x), U01 = sh1(U0A, U0B.x, U0B.y);
+  vf4 Up0 = UpA, Upm1 = shm1f(Upm1s, UpA);
+  vf4 Vm0 = VmA, Vm1 = sh1f(VmA, Vm4);
+  vf4 V0m1 = V0A, V00 = sh0(V0A, V0B.x), V01 = sh1(V0A, V0B.x, V0B.y);
 
   const float rdx = a.rdx, rdy = a.rdy;
 
@@ -737,7 +739,6 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   st4(a.dnh, idx, dnh);
   st4(a.dnu, idx, dnu);
   st4(a.dnv, idx, dnv);
-  vf4 doh = ld4(a.doh, idx), dou = ld4(a.dou, idx), dov = ld4(a.dov, idx);
   st4(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
   st4(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
   st4(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
