@@ -74,3 +74,38 @@ def test_sendrecv_jvp_self():
         y, yt = fwd_ad.unpack_dual(yd)
     assert torch.equal(y, x.reshape(6))
     assert torch.equal(yt, t.reshape(6))
+
+
+def test_compose_with_custom_autograd_function():
+    """Collectives inside user custom Functions (analog of the reference's
+    custom_vjp interplay tests, test_allreduce.py:226-322)."""
+
+    class Scale2(torch.autograd.Function):
+        @staticmethod
+        def forward(ctx, x):
+            return m.allreduce(x.detach() * 2, m.SUM)
+
+        @staticmethod
+        def backward(ctx, g):
+            return m.allreduce(g, m.SUM) * 2
+
+    x = torch.randn(4, requires_grad=True)
+    y = Scale2.apply(x)
+    assert torch.allclose(y, 2 * x)
+    y.sum().backward()
+    assert torch.allclose(x.grad, 2 * torch.ones(4))
+
+
+def test_compose_with_checkpointing():
+    """allreduce under torch.utils.checkpoint: the recompute path re-runs
+    the collective symmetrically on every rank."""
+    from torch.utils.checkpoint import checkpoint
+
+    def block(x):
+        return torch.tanh(m.allreduce(x, m.SUM)) * 3
+
+    x = torch.randn(5, dtype=torch.float64, requires_grad=True)
+    y = checkpoint(block, x, use_reentrant=False)
+    y.sum().backward()
+    expect = 3 * (1 - torch.tanh(x.detach()) ** 2)
+    assert torch.allclose(x.grad, expect, atol=1e-12)
