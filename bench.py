@@ -35,11 +35,13 @@ def measure_allreduce_gbps(steps=20, warmup=5):
     nbytes = 256 * 1024 * 1024
     n = nbytes // 2
     x = torch.randn(n, device="cuda").to(torch.bfloat16)
-    # register the persistent buffer so RCCL can use zero-copy protocols
+    m.allreduce(x, m.SUM)  # creates the default comm's RCCL communicator
+    # register the persistent input so RCCL can use zero-copy protocols
     try:
         from mpi4jax_amd._backend import rccl
+        from mpi4jax_amd.parallel.comm import get_default_comm
 
-        rccl.ext().comm_register(comm.rccl_handle(), x)
+        rccl.ext().comm_register(get_default_comm().rccl_handle(), x)
     except Exception:
         pass
     for _ in range(warmup):
