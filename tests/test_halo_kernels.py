@@ -1,0 +1,113 @@
+"""Direct unit tests for the halo-engine kernels.
+
+These kernels otherwise run only inside *remote* exchanges (multi-GPU),
+so pin their semantics against plain torch indexing here at world 1.
+Layouts must match parallel/grid.py halo_plan:
+pack buffers are [field-major per column] for cols and [diag d][field f]
+for corners with send cells d0 (1,1), d1 (1,nx-2), d2 (ny-2,1),
+d3 (ny-2,nx-2) and recv cells d0 (ny-1,nx-1), d1 (ny-1,0), d2 (0,nx-1),
+d3 (0,0).
+"""
+
+import pytest
+import torch
+
+import mpi4jax_amd as m  # noqa: F401  (init side effects)
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope="module")
+def ext():
+    import mpi4jax_amd._rccl_C as e
+
+    return e
+
+
+def fields(ny=7, nx=9, nf=3, dtype=torch.float32, seed=0):
+    g = torch.Generator().manual_seed(seed)
+    return [torch.randn(ny, nx, generator=g).to(dtype).cuda()
+            for _ in range(nf)]
+
+
+@pytest.mark.parametrize("nf", [1, 2, 3])
+@pytest.mark.parametrize("col", [0, 1, 7, 8])
+def test_pack_cols(ext, nf, col):
+    fs = fields(nf=nf)
+    ny = fs[0].shape[0]
+    buf = torch.empty(nf * ny, device="cuda")
+    ext.pack_cols(buf, fs, col)
+    torch.cuda.synchronize()
+    expect = torch.cat([f[:, col] for f in fs])
+    assert torch.equal(buf, expect)
+
+
+@pytest.mark.parametrize("col", [0, 8])
+def test_unpack_cols(ext, col):
+    fs = fields()
+    ny = fs[0].shape[0]
+    originals = [f.clone() for f in fs]
+    data = torch.randn(3 * ny, device="cuda")
+    ext.unpack_cols(fs, data, col)
+    torch.cuda.synchronize()
+    for i, (f, orig) in enumerate(zip(fs, originals)):
+        assert torch.equal(f[:, col], data[i * ny:(i + 1) * ny])
+        mask = torch.ones_like(f, dtype=torch.bool)
+        mask[:, col] = False
+        assert torch.equal(f[mask], orig[mask])  # nothing else touched
+
+
+def test_pack_corners_layout(ext):
+    fs = fields(nf=3)
+    ny, nx = fs[0].shape
+    buf = torch.empty(12, device="cuda")
+    ext.pack_corners(buf, fs)
+    torch.cuda.synchronize()
+    send_cells = [(1, 1), (1, nx - 2), (ny - 2, 1), (ny - 2, nx - 2)]
+    for d, (j, i) in enumerate(send_cells):
+        for f_idx, f in enumerate(fs):
+            assert buf[d * 3 + f_idx].item() == f[j, i].item(), (d, f_idx)
+
+
+@pytest.mark.parametrize("mask", [0b1111, 0b0101, 0b0010, 0])
+def test_unpack_corners_mask(ext, mask):
+    fs = fields(nf=2)
+    ny, nx = fs[0].shape
+    originals = [f.clone() for f in fs]
+    data = torch.arange(8.0, device="cuda")
+    ext.unpack_corners(fs, data, mask)
+    torch.cuda.synchronize()
+    recv_cells = [(ny - 1, nx - 1), (ny - 1, 0), (0, nx - 1), (0, 0)]
+    for d, (j, i) in enumerate(recv_cells):
+        for f_idx, f in enumerate(fs):
+            if mask & (1 << d):
+                assert f[j, i].item() == data[d * 2 + f_idx].item()
+            else:
+                assert f[j, i].item() == originals[f_idx][j, i].item()
+
+
+def test_halo_wrap_sides(ext):
+    for side in (0, 1, 2):
+        fs = fields(nf=3)
+        originals = [f.clone() for f in fs]
+        ext.halo_wrap(fs, side)
+        torch.cuda.synchronize()
+        for f, orig in zip(fs, originals):
+            if side in (0, 2):
+                assert torch.equal(f[:, -1], orig[:, 1])
+            else:
+                assert torch.equal(f[:, -1], orig[:, -1])
+            if side in (1, 2):
+                assert torch.equal(f[:, 0], orig[:, -2])
+            else:
+                assert torch.equal(f[:, 0], orig[:, 0])
+            assert torch.equal(f[:, 1:-1], orig[:, 1:-1])
+
+
+def test_pack_cols_f64(ext):
+    fs = fields(dtype=torch.float64, nf=2)
+    ny = fs[0].shape[0]
+    buf = torch.empty(2 * ny, dtype=torch.float64, device="cuda")
+    ext.pack_cols(buf, fs, 3)
+    torch.cuda.synchronize()
+    assert torch.equal(buf, torch.cat([f[:, 3] for f in fs]))
