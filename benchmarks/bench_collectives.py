@@ -91,6 +91,27 @@ def bench_bisection(comm, device, steps=10, warmup=3, gib_per_rank=1):
     return recs
 
 
+def bench_reduce_scatter(comm, device, steps=20, warmup=5, mib=256):
+    """reduce_scatter bandwidth (the DP gradient-sharding primitive)."""
+    n = comm.size
+    nbytes = mib * 1024 * 1024
+    elems = nbytes // 2
+    x = torch.randn(elems, device=device).to(torch.bfloat16).reshape(n, -1)
+
+    def fn():
+        m.reduce_scatter(x, m.SUM, comm=comm)
+
+    dt = _timeit(fn, steps, warmup, comm)
+    algbw = nbytes / dt / 1e9
+    busbw = algbw * ((n - 1) / n) if n > 1 else algbw
+    rec = {"bench": "reduce_scatter", "bytes": nbytes, "dtype": "bf16",
+           "n_gpus": n, "time_us": round(dt * 1e6, 1),
+           "algbw_GBps": round(algbw, 2), "busbw_GBps": round(busbw, 2)}
+    if comm.rank == 0:
+        print(json.dumps(rec), flush=True)
+    return rec
+
+
 def bench_grad(comm, device, steps=20, warmup=5, mib=256):
     """Backward through allreduce(SUM): fwd allreduce + identity VJP."""
     nbytes = mib * 1024 * 1024
@@ -115,7 +136,8 @@ def bench_grad(comm, device, steps=20, warmup=5, mib=256):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("which", nargs="?", default="all",
-                   choices=["all", "allreduce", "bisection", "grad"])
+                   choices=["all", "allreduce", "bisection", "grad",
+                            "reduce_scatter"])
     p.add_argument("--gib", type=int, default=1)
     args = p.parse_args()
     m.init()
@@ -125,6 +147,8 @@ def main():
         bench_allreduce(comm, device)
     if args.which in ("all", "bisection"):
         bench_bisection(comm, device, gib_per_rank=args.gib)
+    if args.which in ("all", "reduce_scatter"):
+        bench_reduce_scatter(comm, device)
     if args.which in ("all", "grad"):
         bench_grad(comm, device)
 
