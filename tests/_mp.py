@@ -32,13 +32,28 @@ def _bootstrap(rank, world_size, port, fn, args):
         m.finalize()
 
 
-def run_multiproc(fn, world_size=2, args=(), timeout=180):
-    """Spawn `world_size` processes running fn(rank, world_size, *args)."""
+def run_multiproc(fn, world_size=2, args=(), timeout=240):
+    """Spawn `world_size` processes running fn(rank, world_size, *args).
+
+    Enforces a wall-clock timeout so a deadlocked worker fails the test
+    instead of hanging the whole suite.
+    """
+    import time
+
     port = _free_port()
-    mp.start_processes(
+    ctx = mp.start_processes(
         _bootstrap,
         args=(world_size, port, fn, args),
         nprocs=world_size,
-        join=True,
+        join=False,
         start_method="spawn",
     )
+    deadline = time.time() + timeout
+    while not ctx.join(timeout=5):
+        if time.time() > deadline:
+            for p in ctx.processes:
+                if p.is_alive():
+                    p.terminate()
+            raise TimeoutError(
+                f"multiprocess test exceeded {timeout}s (deadlock?)"
+            )
