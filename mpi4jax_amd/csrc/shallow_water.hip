@@ -380,6 +380,28 @@ __device__ inline void st4(float* p, long long off, vf4 v) {
   *(vf4*)(p + off) = v;
 }
 
+typedef float vf2 __attribute__((ext_vector_type(2), aligned(4)));
+
+__device__ inline vf2 ld2(const float* p, long long off) {
+  return *(const vf2*)(p + off);
+}
+
+// shifted-vector builders: lane c of the result holds value at i0+c+k
+__device__ inline vf4 sh0(vf4 Am1, float x3) {
+  // offset 0 from a load at idx-1 plus the i0+3 element
+  return (vf4){Am1.y, Am1.z, Am1.w, x3};
+}
+__device__ inline vf4 sh1(vf4 Am1, float x3, float x4) {
+  return (vf4){Am1.z, Am1.w, x3, x4};
+}
+__device__ inline vf4 sh0f(vf4 A0) { return A0; }  // load at idx
+__device__ inline vf4 sh1f(vf4 A0, float x4) {
+  return (vf4){A0.y, A0.z, A0.w, x4};
+}
+__device__ inline vf4 shm1f(float xm1, vf4 A0) {
+  return (vf4){xm1, A0.x, A0.y, A0.z};
+}
+
 __device__ inline vf4 rcp4(vf4 x) {
   // v_rcp_f32 per lane (~1 ulp): removes correctly-rounded-division
   // chains (~30 unpipelined cycles each) from the per-pack critical path
@@ -600,10 +622,16 @@ __global__ void sw_stage7v(SwArgs<float> a) {
     return;
   }
   const long long idx = (long long)j * nx + i0;
-  vf4 uc = ld4(a.u, idx), uw = ld4(a.u, idx - 1), ue = ld4(a.u, idx + 1);
+  // 8 loads instead of 10: center/west/east derived from one unaligned
+  // ld4 + ld2 pair per row (same trick as stage18v)
+  vf4 uA = ld4(a.u, idx - 1);
+  vf2 uB = ld2(a.u, idx + 3);
   vf4 un = ld4(a.u, idx + nx), us = ld4(a.u, idx - nx);
-  vf4 vcc = ld4(a.v, idx), vw = ld4(a.v, idx - 1), ve = ld4(a.v, idx + 1);
+  vf4 vA = ld4(a.v, idx - 1);
+  vf2 vB = ld2(a.v, idx + 3);
   vf4 vn = ld4(a.v, idx + nx), vs = ld4(a.v, idx - nx);
+  vf4 uw = uA, uc = sh0(uA, uB.x), ue = sh1(uA, uB.x, uB.y);
+  vf4 vw = vA, vcc = sh0(vA, vB.x), ve = sh1(vA, vB.x, vB.y);
   const float nu = a.nu;
   vf4 lu = (nu * (ue - uc) * a.rdx - nu * (uc - uw) * a.rdx) * a.rdx +
            (nu * (un - uc) * a.rdy - nu * (uc - us) * a.rdy) * a.rdy;
@@ -617,28 +645,6 @@ __global__ void sw_stage7v(SwArgs<float> a) {
 // vector stage 18 = merged stage 8 vectorized: tendencies + update
 // computed straight from h/u/v with float4 packs and shifted vectors —
 // 312 MB/step of HBM traffic instead of the two-pass 598 MB.
-typedef float vf2 __attribute__((ext_vector_type(2), aligned(4)));
-
-__device__ inline vf2 ld2(const float* p, long long off) {
-  return *(const vf2*)(p + off);
-}
-
-// shifted-vector builders: lane c of the result holds value at i0+c+k
-__device__ inline vf4 sh0(vf4 Am1, float x3) {
-  // offset 0 from a load at idx-1 plus the i0+3 element
-  return (vf4){Am1.y, Am1.z, Am1.w, x3};
-}
-__device__ inline vf4 sh1(vf4 Am1, float x3, float x4) {
-  return (vf4){Am1.z, Am1.w, x3, x4};
-}
-__device__ inline vf4 sh0f(vf4 A0) { return A0; }  // load at idx
-__device__ inline vf4 sh1f(vf4 A0, float x4) {
-  return (vf4){A0.y, A0.z, A0.w, x4};
-}
-__device__ inline vf4 shm1f(float xm1, vf4 A0) {
-  return (vf4){xm1, A0.x, A0.y, A0.z};
-}
-
 __global__ void sw_stage18v(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 3) / 4;
