@@ -91,3 +91,18 @@ def test_reduce_gather_scatter_compiled():
 
     x = torch.arange(5.0)
     assert torch.equal(f(x), 3 * x)
+
+
+@pytest.mark.gpu
+def test_compiled_ops_gpu_roundtrip():
+    @torch.compile(fullgraph=True)
+    def f(x):
+        a = jit_ops.scan(x, "sum")
+        b = jit_ops.sendrecv(x, x.detach(), source=0, dest=0)
+        c = jit_ops.reduce_scatter(x[None], "sum")
+        return a + b + c
+
+    x = torch.arange(6.0, device="cuda")
+    y = f(x)
+    torch.cuda.synchronize()
+    assert torch.equal(y, 3 * x)
