@@ -294,6 +294,36 @@ void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
                                cur_stream()));
 }
 
+
+// RCCL's collectives take size_t counts, but point-to-point transfers
+// silently truncate somewhere above 2^31 elements (observed on RCCL
+// 2.26: the tail of an 8.6 GB ncclSend/Recv pair never arrives).  All
+// p2p goes through these chunked wrappers; chunking is deterministic so
+// both sides of a pair split identically inside the group.
+constexpr int64_t kP2PChunk = int64_t(1) << 30;  // elements
+
+void p2p_send(const void* ptr, int64_t count, ncclDataType_t dt,
+              size_t esz, int peer, ncclComm_t comm, hipStream_t stream) {
+  const char* p = (const char*)ptr;
+  while (count > 0) {
+    int64_t c = count < kP2PChunk ? count : kP2PChunk;
+    RCCL_CHECK(ncclSend(p, c, dt, peer, comm, stream));
+    p += c * esz;
+    count -= c;
+  }
+}
+
+void p2p_recv(void* ptr, int64_t count, ncclDataType_t dt, size_t esz,
+              int peer, ncclComm_t comm, hipStream_t stream) {
+  char* p = (char*)ptr;
+  while (count > 0) {
+    int64_t c = count < kP2PChunk ? count : kP2PChunk;
+    RCCL_CHECK(ncclRecv(p, c, dt, peer, comm, stream));
+    p += c * esz;
+    count -= c;
+  }
+}
+
 // grouped p2p composition: RCCL has no alltoall/gather/scatter primitives
 //  — on the fully-connected xGMI clique direct per-peer
 // send/recv IS the bandwidth-optimal algorithm (every peer pair has its own
@@ -312,10 +342,10 @@ void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
   log_enqueue("Alltoall", c, in.numel());
   RCCL_CHECK(ncclGroupStart());
   for (int r = 0; r < c.size; ++r) {
-    RCCL_CHECK(ncclSend(ip + r * chunk * esz, chunk, dt, r, c.comm,
-                        cur_stream()));
-    RCCL_CHECK(ncclRecv(op_ + r * chunk * esz, chunk, dt, r, c.comm,
-                        cur_stream()));
+    p2p_send(ip + r * chunk * esz, chunk, dt, esz, r, c.comm,
+             cur_stream());
+    p2p_recv(op_ + r * chunk * esz, chunk, dt, esz, r, c.comm,
+             cur_stream());
   }
   RCCL_CHECK(ncclGroupEnd());
 }
@@ -329,14 +359,14 @@ void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   auto dt = nccl_dtype(in);
   log_enqueue("Gather", c, chunk);
   RCCL_CHECK(ncclGroupStart());
-  RCCL_CHECK(ncclSend(in.data_ptr(), chunk, dt, (int)root, c.comm,
-                      cur_stream()));
+  p2p_send(in.data_ptr(), chunk, dt, esz, (int)root, c.comm,
+           cur_stream());
   if (c.rank == (int)root) {
     TORCH_CHECK(out.numel() == chunk * c.size, "gather out size mismatch");
     char* op_ = (char*)out.data_ptr();
     for (int r = 0; r < c.size; ++r) {
-      RCCL_CHECK(ncclRecv(op_ + r * chunk * esz, chunk, dt, r, c.comm,
-                          cur_stream()));
+      p2p_recv(op_ + r * chunk * esz, chunk, dt, esz, r, c.comm,
+               cur_stream());
     }
   }
   RCCL_CHECK(ncclGroupEnd());
@@ -355,12 +385,12 @@ void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
     TORCH_CHECK(in.numel() == chunk * c.size, "scatter in size mismatch");
     char* ip = (char*)in.data_ptr();
     for (int r = 0; r < c.size; ++r) {
-      RCCL_CHECK(ncclSend(ip + r * chunk * esz, chunk, dt, r, c.comm,
-                          cur_stream()));
+      p2p_send(ip + r * chunk * esz, chunk, dt, esz, r, c.comm,
+               cur_stream());
     }
   }
-  RCCL_CHECK(ncclRecv(out.data_ptr(), chunk, dt, (int)root, c.comm,
-                      cur_stream()));
+  p2p_recv(out.data_ptr(), chunk, dt, esz, (int)root, c.comm,
+           cur_stream());
   RCCL_CHECK(ncclGroupEnd());
 }
 
@@ -369,8 +399,8 @@ void send(at::Tensor in, int64_t dest, int64_t comm_id) {
   auto c = get_comm(comm_id);
   TORCH_CHECK(in.is_cuda() && in.is_contiguous(), "bad send input");
   log_enqueue("Send", c, in.numel());
-  RCCL_CHECK(ncclSend(in.data_ptr(), in.numel(), nccl_dtype(in), (int)dest,
-                      c.comm, cur_stream()));
+  p2p_send(in.data_ptr(), in.numel(), nccl_dtype(in), in.element_size(),
+           (int)dest, c.comm, cur_stream());
 }
 
 void recv(at::Tensor out, int64_t source, int64_t comm_id) {
@@ -378,8 +408,8 @@ void recv(at::Tensor out, int64_t source, int64_t comm_id) {
   auto c = get_comm(comm_id);
   TORCH_CHECK(out.is_cuda() && out.is_contiguous(), "bad recv output");
   log_enqueue("Recv", c, out.numel());
-  RCCL_CHECK(ncclRecv(out.data_ptr(), out.numel(), nccl_dtype(out),
-                      (int)source, c.comm, cur_stream()));
+  p2p_recv(out.data_ptr(), out.numel(), nccl_dtype(out),
+           out.element_size(), (int)source, c.comm, cur_stream());
 }
 
 void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
@@ -391,11 +421,10 @@ void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
   log_enqueue("Sendrecv", c, sendbuf.numel());
   // grouped => deadlock-free by construction
   RCCL_CHECK(ncclGroupStart());
-  RCCL_CHECK(ncclSend(sendbuf.data_ptr(), sendbuf.numel(),
-                      nccl_dtype(sendbuf), (int)dest, c.comm, cur_stream()));
-  RCCL_CHECK(ncclRecv(recvbuf.data_ptr(), recvbuf.numel(),
-                      nccl_dtype(recvbuf), (int)source, c.comm,
-                      cur_stream()));
+  p2p_send(sendbuf.data_ptr(), sendbuf.numel(), nccl_dtype(sendbuf),
+           sendbuf.element_size(), (int)dest, c.comm, cur_stream());
+  p2p_recv(recvbuf.data_ptr(), recvbuf.numel(), nccl_dtype(recvbuf),
+           recvbuf.element_size(), (int)source, c.comm, cur_stream());
   RCCL_CHECK(ncclGroupEnd());
 }
 
@@ -425,14 +454,14 @@ void scan(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
                              hipMemcpyDeviceToDevice, stream));
   } else {
     // receive the prefix of ranks [0, r) into out, then out = out (+) in
-    RCCL_CHECK(ncclRecv(out.data_ptr(), out.numel(), dt, c.rank - 1, c.comm,
-                        stream));
+    p2p_recv(out.data_ptr(), out.numel(), dt, out.element_size(),
+             c.rank - 1, c.comm, stream);
     launch_combine(out.data_ptr(), out.data_ptr(), in.data_ptr(),
                    in.numel(), dt_code(in), (int)op, stream);
   }
   if (c.rank < c.size - 1) {
-    RCCL_CHECK(ncclSend(out.data_ptr(), out.numel(), dt, c.rank + 1, c.comm,
-                        stream));
+    p2p_send(out.data_ptr(), out.numel(), dt, out.element_size(),
+             c.rank + 1, c.comm, stream);
   }
   log_enqueue("Scan", c, in.numel());
 }
