@@ -172,6 +172,11 @@ class CartesianGrid:
         from ..ops.recv import recv
 
         me = self.comm.rank
+        if out.dim() != 2 or out.shape[0] < 3 or out.shape[1] < 3:
+            raise ValueError(
+                f"halo_exchange needs a 2-D array of at least 3x3 "
+                f"(1-cell halo ring + interior), got {tuple(out.shape)}"
+            )
         ny, nx = out.shape
 
         def xchg(send_to, recv_from, send_view, recv_setter, template):

@@ -38,3 +38,11 @@ def test_halo_exchange_self_periodic():
 def test_bad_grid():
     with pytest.raises(ValueError):
         CartesianGrid(m.COMM_WORLD, dims=(2, 2))
+
+
+def test_halo_exchange_rejects_degenerate():
+    g = CartesianGrid(m.COMM_WORLD, dims=(1, 1), periodic=(False, True))
+    with pytest.raises(ValueError, match="3x3"):
+        g.halo_exchange(torch.zeros(2, 5))
+    with pytest.raises(ValueError, match="3x3"):
+        g.halo_exchange(torch.zeros(4))
