@@ -404,3 +404,18 @@ def test_huge_collectives_64bit_counts():
     assert sc[-1].item() == 1.0
     del sc, x
     torch.cuda.empty_cache()
+
+
+def test_comm_lifecycle_many():
+    """Clone/free cycles must not leak registry entries."""
+    import mpi4jax_amd._rccl_C as ext
+
+    world = m.get_world()
+    base = ext.comm_count()
+    for _ in range(6):
+        c = world.Clone()
+        y = m.allreduce(torch.ones(64, device="cuda"), m.SUM, comm=c)
+        torch.cuda.synchronize()
+        assert y[0].item() == 1.0
+        c.free()
+    assert ext.comm_count() == base
