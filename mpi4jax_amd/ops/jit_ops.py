@@ -17,7 +17,9 @@ Usage::
 
 Communicators cross the graph boundary as int registry keys (the same
 int64-handle marshalling idea the reference uses for MPI objects,
-``mpi_ops_common.h:36-48``).
+``mpi_ops_common.h:36-48``): pass ``comm=None`` for the default, or
+``comm=jit_ops.comm_key(my_comm)`` computed outside the compiled region —
+the key is runtime data, so one graph serves any communicator.
 
 Not traced: ``send``/``recv`` (their buffered self-messaging queue is
 process state a traced graph must not capture) and ``barrier`` (returns
@@ -36,14 +38,14 @@ _COMMS = {}
 _KEYS = {}
 
 
-@torch._dynamo.assume_constant_result
 def comm_key(comm) -> int:
     """Registry key for a communicator.
 
-    Marked constant for dynamo: at trace time it runs eagerly (communicator
-    creation is not traceable) and the key is baked into the graph — dynamo
-    guards on the ``comm`` argument, so a different communicator object
-    triggers a recompile with its own key.
+    Inside a ``torch.compile(fullgraph=True)`` region pass either
+    ``comm=None`` (the default communicator, resolved once at trace time)
+    or this function's result computed OUTSIDE the graph — int keys flow
+    through the graph as runtime data, so one compiled function serves any
+    number of communicators without recompilation.
     """
     comm = resolve_comm(comm)
     if id(comm) not in _KEYS:
@@ -51,6 +53,21 @@ def comm_key(comm) -> int:
         _COMMS[key] = comm
         _KEYS[id(comm)] = key
     return _KEYS[id(comm)]
+
+
+@torch._dynamo.assume_constant_result
+def _default_key() -> int:
+    # zero-arg: safe for assume_constant_result; runs eagerly at trace
+    # time (communicator creation is untraceable), baked per-graph
+    return comm_key(None)
+
+
+def _key_of(comm) -> int:
+    if comm is None:
+        return _default_key()
+    if isinstance(comm, int):
+        return comm
+    return comm_key(comm)
 
 
 def _comm(key: int) -> Communicator:
@@ -88,7 +105,7 @@ torch.library.register_autograd("mpi4jax_amd::allreduce", _allreduce_bwd,
 
 def allreduce(x, op=Op.SUM, *, comm=None):
     op = op.value if isinstance(op, Op) else str(op)
-    return _allreduce(x, op, comm_key(comm))
+    return _allreduce(x, op, _key_of(comm))
 
 
 # ----------------------------------------------------------------- allgather
@@ -105,7 +122,7 @@ def _(x, key):
 
 
 def allgather(x, *, comm=None):
-    return _allgather(x, comm_key(comm))
+    return _allgather(x, _key_of(comm))
 
 
 # ----------------------------------------------------------------- alltoall
@@ -121,7 +138,7 @@ def _(x, key):
 
 
 def alltoall(x, *, comm=None):
-    return _alltoall(x, comm_key(comm))
+    return _alltoall(x, _key_of(comm))
 
 
 # ------------------------------------------------------------ reduce_scatter
@@ -139,7 +156,7 @@ def _(x, op, key):
 
 def reduce_scatter(x, op=Op.SUM, *, comm=None):
     op = op.value if isinstance(op, Op) else str(op)
-    return _reduce_scatter(x, op, comm_key(comm))
+    return _reduce_scatter(x, op, _key_of(comm))
 
 
 # ----------------------------------------------------------------- bcast
@@ -155,7 +172,7 @@ def _(x, root, key):
 
 
 def bcast(x, root, *, comm=None):
-    return _bcast(x, root, comm_key(comm))
+    return _bcast(x, root, _key_of(comm))
 
 
 # ----------------------------------------------------------------- scan
@@ -172,7 +189,7 @@ def _(x, op, key):
 
 def scan(x, op=Op.SUM, *, comm=None):
     op = op.value if isinstance(op, Op) else str(op)
-    return _scan(x, op, comm_key(comm))
+    return _scan(x, op, _key_of(comm))
 
 
 # ----------------------------------------------------------------- sendrecv
@@ -210,7 +227,7 @@ torch.library.register_autograd("mpi4jax_amd::sendrecv", _sendrecv_bwd,
 
 
 def sendrecv(sendbuf, recvbuf, source, dest, *, comm=None):
-    return _sendrecv(sendbuf, recvbuf, source, dest, comm_key(comm))
+    return _sendrecv(sendbuf, recvbuf, source, dest, _key_of(comm))
 
 
 # ----------------------------------------------------------------- reduce
@@ -228,7 +245,7 @@ def _(x, op, root, key):
 
 def reduce(x, op, root, *, comm=None):
     op = op.value if isinstance(op, Op) else str(op)
-    return _reduce(x, op, root, comm_key(comm))
+    return _reduce(x, op, root, _key_of(comm))
 
 
 # ----------------------------------------------------------------- gather
@@ -250,7 +267,7 @@ def gather(x, root, *, comm=None):
     """Root gets ``(nproc, *shape)``; other ranks get an undefined tensor
     of that shape (static shapes are required inside a compiled graph —
     the eager op's non-root input passthrough does not translate)."""
-    return _gather(x, root, comm_key(comm))
+    return _gather(x, root, _key_of(comm))
 
 
 # ----------------------------------------------------------------- scatter
@@ -268,4 +285,4 @@ def _(x, root, key):
 
 
 def scatter(x, root, *, comm=None):
-    return _scatter(x, root, comm_key(comm))
+    return _scatter(x, root, _key_of(comm))

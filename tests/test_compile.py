@@ -106,3 +106,29 @@ def test_compiled_ops_gpu_roundtrip():
     y = f(x)
     torch.cuda.synchronize()
     assert torch.equal(y, 3 * x)
+
+
+def test_comm_key_recompiles_per_communicator(monkeypatch):
+    """Communicators cross compiled graphs as int keys computed OUTSIDE
+    the graph; keys are runtime data, so one compiled function serves any
+    communicator with the correct registry entry at runtime."""
+    import mpi4jax_amd as m
+    from mpi4jax_amd.ops import jit_ops as jo
+
+    seen = []
+    orig = jo._comm
+    monkeypatch.setattr(jo, "_comm", lambda k: (seen.append(k), orig(k))[1])
+
+    ca = m.get_world().Clone()
+    cb = m.get_world().Clone()
+    ka, kb = jo.comm_key(ca), jo.comm_key(cb)
+
+    @torch.compile(fullgraph=True)
+    def f(x, k):
+        return jo.allreduce(x, "sum", comm=k)
+
+    x = torch.ones(4)
+    f(x, ka)
+    f(x, kb)
+    assert len(seen) >= 2
+    assert seen[-1] == kb and seen[0] == ka, (seen, ka, kb)
