@@ -132,3 +132,14 @@ def test_comm_key_recompiles_per_communicator(monkeypatch):
     f(x, kb)
     assert len(seen) >= 2
     assert seen[-1] == kb and seen[0] == ka, (seen, ka, kb)
+
+
+def test_eager_ops_inside_compile_graph_break_ok():
+    """The eager API inside torch.compile (without fullgraph) must still
+    be correct — dynamo graph-breaks around it and falls back to eager."""
+    @torch.compile  # no fullgraph: breaks are allowed
+    def f(x):
+        return m.allreduce(x * 2, m.SUM) + 1
+
+    x = torch.arange(4.0)
+    assert torch.equal(f(x), 2 * x + 1)
