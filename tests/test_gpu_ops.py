@@ -419,3 +419,15 @@ def test_comm_lifecycle_many():
         assert y[0].item() == 1.0
         c.free()
     assert ext.comm_count() == base
+
+
+def test_ops_on_side_stream():
+    """Collectives enqueue on the CURRENT torch stream, side streams
+    included (c10::hip::getCurrentHIPStream plumbing)."""
+    s = torch.cuda.Stream()
+    x = torch.ones(1 << 20, device="cuda")
+    with torch.cuda.stream(s):
+        y = m.allreduce(x * 3, m.SUM)
+        z = m.sendrecv(y, y, source=0, dest=0)
+    s.synchronize()
+    assert z[0].item() == 3.0 and z[-1].item() == 3.0
