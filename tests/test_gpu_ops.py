@@ -431,3 +431,10 @@ def test_ops_on_side_stream():
         z = m.sendrecv(y, y, source=0, dest=0)
     s.synchronize()
     assert z[0].item() == 3.0 and z[-1].item() == 3.0
+
+
+def test_noncontiguous_inputs_gpu(x):
+    t = x.t()
+    y = m.allreduce(t, m.SUM)
+    torch.cuda.synchronize()
+    assert y.shape == t.shape and torch.equal(y, t)

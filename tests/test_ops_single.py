@@ -142,3 +142,14 @@ def test_batched_collective_idiom():
     stacked = m.allreduce(xs, m.SUM)
     looped = torch.stack([m.allreduce(x, m.SUM) for x in xs])
     assert torch.equal(stacked, looped)
+
+
+def test_noncontiguous_inputs():
+    a = torch.arange(12.0).reshape(3, 4)
+    t = a.t()  # non-contiguous view
+    y = m.allreduce(t, m.SUM)
+    assert y.shape == (4, 3) and torch.equal(y, t)
+    g = m.allgather(t)
+    assert g.shape == (1, 4, 3) and torch.equal(g[0], t)
+    s = m.sendrecv(t, t, source=0, dest=0)
+    assert torch.equal(s, t)
