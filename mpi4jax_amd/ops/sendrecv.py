@@ -82,6 +82,11 @@ def sendrecv(sendbuf, recvbuf, source, dest, *, sendtag=0, recvtag=ANY_TAG,
     sendbuf, comm, backend = prepare(sendbuf, comm, "sendrecv")
     recvbuf = as_tensor(recvbuf, "sendrecv")
     check_dtype(recvbuf, "sendrecv")
+    if recvbuf.device != sendbuf.device:
+        raise ValueError(
+            f"sendrecv: sendbuf ({sendbuf.device}) and recvbuf template "
+            f"({recvbuf.device}) must live on the same device"
+        )
     for name, r in (("source", source), ("dest", dest)):
         if not 0 <= r < comm.size:
             raise ValueError(

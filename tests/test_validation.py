@@ -46,3 +46,13 @@ def test_unsupported_dtype():
 def test_scan_avg_rejected():
     with pytest.raises(ValueError):
         m.scan(torch.zeros(2), m.AVG)
+
+
+def test_sendrecv_device_mismatch():
+    if not torch.cuda.is_available():
+        import pytest as _pytest
+
+        _pytest.skip("needs a GPU to construct the mismatch")
+    with pytest.raises(ValueError, match="same device"):
+        m.sendrecv(torch.zeros(3), torch.zeros(3, device="cuda"),
+                   source=0, dest=0)
