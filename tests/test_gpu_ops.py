@@ -438,3 +438,29 @@ def test_noncontiguous_inputs_gpu(x):
     y = m.allreduce(t, m.SUM)
     torch.cuda.synchronize()
     assert y.shape == t.shape and torch.equal(y, t)
+
+
+def test_twopass_vector_stages_match_default(monkeypatch):
+    """The two-pass vector stages (11+16) are the fallback lineage of the
+    merged stage 18 — keep them equivalent."""
+    from mpi4jax_amd.models import ShallowWater
+
+    results = {}
+    for twopass in ("1", ""):
+        if twopass:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_TWOPASS", "1")
+        else:
+            monkeypatch.delenv("MPI4JAX_AMD_SW_TWOPASS", raising=False)
+        sw = ShallowWater(nx=96, ny=48, device="cuda",
+                          comm=m.get_world().Clone())
+        st = sw.initial_conditions()
+        st = sw.step(st, first_step=True)
+        for _ in range(8):
+            st = sw.step(st)
+        torch.cuda.synchronize()
+        results[twopass] = st
+    for name in ("h", "u", "v"):
+        a, b = getattr(results["1"], name), getattr(results[""], name)
+        assert torch.allclose(a, b, atol=1e-4, rtol=1e-5), (
+            name, (a - b).abs().max().item()
+        )
