@@ -39,6 +39,8 @@ def has_rccl_support() -> bool:
 # (`mpi4jax.has_cuda_support`, utils.py:159-174) — for us "GPU support"
 # means the RCCL extension.
 def has_cuda_support() -> bool:
+    """Alias of :func:`has_rccl_support` (drop-in parity with the
+    reference's probe name, utils.py:159-166)."""
     return has_rccl_support()
 
 

@@ -22,6 +22,8 @@ _LOGGING = False
 
 
 def set_logging(enabled: bool):
+    """Toggle debug logging on every backend (reference:
+    xla_bridge/__init__.py:114-129)."""
     global _LOGGING
     _LOGGING = bool(enabled)
     # keep the native extension's flag in sync if it is loaded
@@ -35,6 +37,7 @@ def set_logging(enabled: bool):
 
 
 def get_logging() -> bool:
+    """Current debug-logging state."""
     return _LOGGING
 
 
