@@ -21,6 +21,7 @@
 
 #include <rocprofiler-sdk-roctx/roctx.h>
 
+#include <algorithm>
 #include <cstdio>
 #include <mutex>
 #include <string>
@@ -167,7 +168,7 @@ void comm_destroy(int64_t id) {
   auto it = g_comms.find(id);
   if (it == g_comms.end()) return;
   ncclCommDestroy(it->second.comm);
-  hipFree(it->second.barrier_buf);
+  (void)hipFree(it->second.barrier_buf);
   g_comms.erase(it);
 }
 
@@ -175,7 +176,7 @@ void destroy_all_comms() {
   std::lock_guard<std::mutex> lk(g_mutex);
   for (auto& kv : g_comms) {
     ncclCommDestroy(kv.second.comm);
-    hipFree(kv.second.barrier_buf);
+    (void)hipFree(kv.second.barrier_buf);
   }
   g_comms.clear();
 }
@@ -205,7 +206,7 @@ void comm_abort(int64_t id) {
   auto it = g_comms.find(id);
   if (it == g_comms.end()) return;
   ncclCommAbort(it->second.comm);
-  hipFree(it->second.barrier_buf);
+  (void)hipFree(it->second.barrier_buf);
   g_comms.erase(it);
 }
 
@@ -233,7 +234,7 @@ py::dict version_info() {
   ncclGetVersion(&nccl_ver);
   d["rccl"] = nccl_ver;
   int hip_ver = 0;
-  hipRuntimeGetVersion(&hip_ver);
+  (void)hipRuntimeGetVersion(&hip_ver);
   d["hip_runtime"] = hip_ver;
   return d;
 }
@@ -611,6 +612,135 @@ void unpack_corners(std::vector<at::Tensor> fields, at::Tensor buf,
                         (int)mask, is_double ? 1 : 0, cur_stream());
 }
 
+// One-call halo-exchange executor for the fused shallow-water step.
+// Executes the schedule computed by parallel/grid.halo_exchange_schedule
+// on the host side once per model (models/shallow_water.py caches it);
+// the wire protocol (message set + enqueue order) is byte-identical to
+// the Python executor `_exchange_fields_py`, whose cross-rank matching
+// is verified per topology in tests/test_shallow_water.py.  Collapses
+// ~30 Python->C++ crossings per exchange into one, which is what keeps
+// the non-graph multi-rank step host-bound-free under strong scaling.
+//
+// Flattened schedule encoding (-1 = no peer):
+//   col_ops: (k, send_to, recv_from, send_col, recv_col) x n
+//   row_ops: (send_to, recv_from, recv_row, send_row) x n
+//   cor_ops: (d, send_to, recv_from) x n
+void sw_exchange(std::vector<at::Tensor> fields,
+                 std::vector<int64_t> wrap_sides,
+                 std::vector<int64_t> col_ops, std::vector<int64_t> row_ops,
+                 std::vector<int64_t> cor_ops, int64_t cor_mask,
+                 std::vector<at::Tensor> col_bufs, at::Tensor cor_sbuf,
+                 at::Tensor cor_rbuf, int64_t comm_id) {
+  ROCTX_SCOPE("mpi4jax_amd::sw_exchange");
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  const int nf = (int)fields.size();
+  TORCH_CHECK(col_ops.size() % 5 == 0 && row_ops.size() % 4 == 0 &&
+                  cor_ops.size() % 3 == 0,
+              "bad sw_exchange schedule");
+  hipStream_t stream = cur_stream();
+
+  // kernel-side periodic wraps (never remote)
+  for (int64_t side : wrap_sides) {
+    launch_halo_wrap(ptrs, nf, ny, nx, (int)side, is_double ? 1 : 0,
+                     stream);
+  }
+  // pack sends
+  for (size_t i = 0; i < col_ops.size(); i += 5) {
+    if (col_ops[i + 1] < 0) continue;  // no send peer
+    const at::Tensor& sb = col_bufs.at(2 * col_ops[i]);
+    TORCH_CHECK(sb.is_cuda() && sb.is_contiguous() &&
+                    sb.numel() >= (int64_t)nf * ny,
+                "bad column send buffer");
+    launch_pack_cols(sb.data_ptr(), ptrs, nf, ny, nx, col_ops[i + 3],
+                     is_double ? 1 : 0, stream);
+  }
+  if (!cor_ops.empty()) {
+    TORCH_CHECK(cor_sbuf.is_cuda() && cor_sbuf.is_contiguous() &&
+                    cor_sbuf.numel() >= 4 * nf,
+                "bad corner send buffer");
+    launch_pack_corners(cor_sbuf.data_ptr(), ptrs, nf, ny, nx,
+                        is_double ? 1 : 0, stream);
+  }
+  bool any_remote = false;
+  for (size_t i = 0; i < col_ops.size(); i += 5)
+    any_remote |= col_ops[i + 1] >= 0 || col_ops[i + 2] >= 0;
+  for (size_t i = 0; i < row_ops.size(); i += 4)
+    any_remote |= row_ops[i] >= 0 || row_ops[i + 1] >= 0;
+  for (size_t i = 0; i < cor_ops.size(); i += 3)
+    any_remote |= cor_ops[i + 1] >= 0 || cor_ops[i + 2] >= 0;
+  if (!any_remote) return;
+
+  auto c = get_comm(comm_id);
+  auto dt = nccl_dtype(fields[0]);
+  const int64_t esz = fields[0].element_size();
+  log_enqueue("SwExchange", c, (int64_t)nf * ny);
+  RCCL_CHECK(ncclGroupStart());
+  for (size_t i = 0; i < col_ops.size(); i += 5) {
+    int64_t k = col_ops[i], st = col_ops[i + 1], rf = col_ops[i + 2];
+    if (st >= 0) {
+      p2p_send(col_bufs.at(2 * k).data_ptr(), (int64_t)nf * ny, dt, esz,
+               (int)st, c.comm, stream);
+    }
+    if (rf >= 0) {
+      const at::Tensor& rb = col_bufs.at(2 * k + 1);
+      TORCH_CHECK(rb.is_cuda() && rb.is_contiguous() &&
+                      rb.numel() >= (int64_t)nf * ny,
+                  "bad column recv buffer");
+      p2p_recv(rb.data_ptr(), (int64_t)nf * ny, dt, esz, (int)rf, c.comm,
+               stream);
+    }
+  }
+  for (size_t i = 0; i < row_ops.size(); i += 4) {
+    int64_t st = row_ops[i], rf = row_ops[i + 1];
+    int64_t ridx = row_ops[i + 2], sidx = row_ops[i + 3];
+    TORCH_CHECK(0 <= std::min(ridx, sidx) && std::max(ridx, sidx) < ny,
+                "bad row index");
+    for (int f = 0; f < nf; ++f) {
+      char* base = (char*)ptrs[f];
+      if (st >= 0) {
+        p2p_send(base + (sidx * nx + 1) * esz, nx - 2, dt, esz, (int)st,
+                 c.comm, stream);
+      }
+      if (rf >= 0) {
+        p2p_recv(base + (ridx * nx + 1) * esz, nx - 2, dt, esz, (int)rf,
+                 c.comm, stream);
+      }
+    }
+  }
+  for (size_t i = 0; i < cor_ops.size(); i += 3) {
+    int64_t d = cor_ops[i], st = cor_ops[i + 1], rf = cor_ops[i + 2];
+    TORCH_CHECK(0 <= d && d < 4, "bad corner index");
+    if (st >= 0) {
+      p2p_send((char*)cor_sbuf.data_ptr() + d * nf * esz, nf, dt, esz,
+               (int)st, c.comm, stream);
+    }
+    if (rf >= 0) {
+      TORCH_CHECK(cor_rbuf.is_cuda() && cor_rbuf.is_contiguous() &&
+                      cor_rbuf.numel() >= 4 * nf,
+                  "bad corner recv buffer");
+      p2p_recv((char*)cor_rbuf.data_ptr() + d * nf * esz, nf, dt, esz,
+               (int)rf, c.comm, stream);
+    }
+  }
+  RCCL_CHECK(ncclGroupEnd());
+
+  // unpack receives (column writes land before corner writes — the one
+  // ordering requirement of halo_plan; stream order provides it)
+  for (size_t i = 0; i < col_ops.size(); i += 5) {
+    if (col_ops[i + 2] < 0) continue;
+    launch_unpack_cols(ptrs, col_bufs.at(2 * col_ops[i] + 1).data_ptr(),
+                       nf, ny, nx, col_ops[i + 4], is_double ? 1 : 0,
+                       stream);
+  }
+  if (cor_mask) {
+    launch_unpack_corners(ptrs, cor_rbuf.data_ptr(), nf, ny, nx,
+                          (int)cor_mask, is_double ? 1 : 0, stream);
+  }
+}
+
 // direct access to the combine kernel (used by gpu numerics tests)
 void combine(at::Tensor dst, at::Tensor a, at::Tensor b, int64_t op) {
   check_pair(dst, a);
@@ -658,4 +788,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("unpack_cols", &unpack_cols);
   m.def("pack_corners", &pack_corners);
   m.def("unpack_corners", &unpack_corners);
+  m.def("sw_exchange", &sw_exchange);
 }

@@ -26,8 +26,7 @@ from collections import namedtuple
 import torch
 
 from ..parallel.comm import resolve_comm
-from ..parallel.grid import (CartesianGrid, halo_plan,
-                             halo_exchange_schedule)
+from ..parallel.grid import CartesianGrid, halo_exchange_schedule
 
 ModelState = namedtuple("ModelState", "h u v dh du dv")
 
@@ -223,6 +222,28 @@ class ShallowWater:
         for k in names:
             fb[k], fb[f"{k}_alt"] = fb[f"{k}_alt"], fb[k]
 
+    def _stage_plan(self):
+        """Kernel-id triple for the (pre-tendency, tendency, friction)
+        stages — static per (dtype, env) so it is resolved once at buffer
+        init, not per step.
+
+        float32 runs the vectorized (float4) stage kernels; f64 and the
+        MPI4JAX_AMD_SW_NOVEC escape hatch run the scalar ones.  Stage 8
+        (fully merged, derived fields in-register) measured slower than
+        the two-pass kernels — kept for reference, selectable via
+        MPI4JAX_AMD_SW_MERGED.
+        """
+        import os
+
+        if os.environ.get("MPI4JAX_AMD_SW_MERGED"):
+            return None, 8, 7
+        if (self.dtype == torch.float32
+                and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
+            if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
+                return 11, 16, 17
+            return None, 18, 17  # merged+vectorized single pass
+        return 1, 6, 7
+
     def _step_fused(self, state, first_step=False):
         from .._backend import rccl
 
@@ -231,9 +252,16 @@ class ShallowWater:
         if fb is None or state.h is not fb["h"]:
             self._init_fused_buffers(state)
             fb = self._fb
-        flags = self._fused_flags()
-        cor_base = float(CORIOLIS_F + float(self.y_local[0]) * CORIOLIS_BETA)
-        cor_dj = float(self.dy * CORIOLIS_BETA)
+        # step-invariant host work is resolved once per buffer (re)init
+        const = fb.get("step_const")
+        if const is None:
+            const = fb["step_const"] = (
+                self._fused_flags(),
+                float(CORIOLIS_F + float(self.y_local[0]) * CORIOLIS_BETA),
+                float(self.dy * CORIOLIS_BETA),
+                self._stage_plan(),
+            )
+        flags, cor_base, cor_dj, (s1, s6, s7) = const
         ab_a, ab_b = ((1.0, 0.0) if first_step
                       else (ADAMS_BASHFORTH_A, ADAMS_BASHFORTH_B))
 
@@ -246,23 +274,6 @@ class ShallowWater:
                          self.lateral_viscosity, cor_base, cor_dj, ab_a,
                          ab_b, flags)
 
-        import os
-
-        # float32 runs the vectorized (float4) stage kernels; f64 and the
-        # MPI4JAX_AMD_SW_NOVEC escape hatch run the scalar ones.  Stage 8
-        # (fully merged, derived fields in-register) measured slower than
-        # the two-pass kernels — kept for reference, selectable via
-        # MPI4JAX_AMD_SW_MERGED.
-        if os.environ.get("MPI4JAX_AMD_SW_MERGED"):
-            s1, s6, s7 = None, 8, 7
-        elif (self.dtype == torch.float32
-                and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
-            if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
-                s1, s6, s7 = 11, 16, 17
-            else:
-                s1, s6, s7 = None, 18, 17  # merged+vectorized single pass
-        else:
-            s1, s6, s7 = 1, 6, 7
         if s1 is not None:
             stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
         stage(s6)         # tendencies + time update -> h_alt/u_alt/v_alt
@@ -279,31 +290,77 @@ class ShallowWater:
                           fb["do_v"])
 
     # ------------------------------------------------------------------
+    def _exchange_cache(self):
+        """Step-invariant halo-exchange state: the resolved schedule (as
+        flat int lists for the one-call C++ executor), the staging
+        buffers, and the RCCL handle.  Built once per fused-buffer init.
+        """
+        fb = self._fb
+        cache = fb.get("xcache")
+        if cache is not None:
+            return cache
+        ny, nx = self.ny_local, self.nx_local
+        for k in ("col_sbuf0", "col_rbuf0", "col_sbuf1", "col_rbuf1"):
+            fb[k] = torch.empty(3 * ny, dtype=self.dtype,
+                                device=self.device)
+        fb["cor_sbuf"] = torch.empty(12, dtype=self.dtype,
+                                     device=self.device)
+        fb["cor_rbuf"] = torch.empty_like(fb["cor_sbuf"])
+        (wrap_sides, col_ops, row_ops, cor_ops,
+         cor_mask) = halo_exchange_schedule(self.grid, nx, ny)
+        z = -1  # C++ encoding of "no peer"
+
+        def flat(ops):
+            return [z if v is None else int(v) for op in ops for v in op]
+
+        has_remote = bool(col_ops or row_ops or cor_ops)
+        comm_id = self.comm.rccl_handle() if has_remote else -1
+        cache = fb["xcache"] = {
+            "sched": (wrap_sides, col_ops, row_ops, cor_ops, cor_mask),
+            "flat": (list(wrap_sides), flat(col_ops), flat(row_ops),
+                     flat(cor_ops), cor_mask),
+            "col_bufs": [fb["col_sbuf0"], fb["col_rbuf0"],
+                         fb["col_sbuf1"], fb["col_rbuf1"]],
+            "comm_id": comm_id,
+        }
+        return cache
+
     def _exchange_fields(self, fields):
         """One-group halo exchange (halo_plan is order-independent).
 
         Self-wrap columns run as a kernel (no RCCL — keeps the world-1
         path graph-capturable with zero comm init); everything remote —
         packed columns, in-place interior rows, diagonal corners — goes
-        into a SINGLE RCCL group enqueue.
+        into a SINGLE RCCL group enqueue.  The whole exchange is one
+        C++ call (``sw_exchange``) so the non-graph multi-rank loop is
+        not host-bound; MPI4JAX_AMD_SW_PYEXCHANGE selects the
+        wire-identical per-op Python executor instead.
         """
+        import os
+
+        from .._backend import rccl
+
+        cache = self._exchange_cache()
+        if os.environ.get("MPI4JAX_AMD_SW_PYEXCHANGE"):
+            return self._exchange_fields_py(fields, cache)
+        fb = self._fb
+        wrap_sides, col_flat, row_flat, cor_flat, cor_mask = cache["flat"]
+        rccl.ext().sw_exchange(
+            fields, wrap_sides, col_flat, row_flat, cor_flat, cor_mask,
+            cache["col_bufs"], fb["cor_sbuf"], fb["cor_rbuf"],
+            cache["comm_id"],
+        )
+
+    def _exchange_fields_py(self, fields, cache):
+        """Per-op executor for the halo schedule (reference/debugging —
+        identical message set and enqueue order to ``sw_exchange``)."""
         from .._backend import rccl
 
         ext = rccl.ext()
-        g = self.grid
-        me = self.comm.rank
         nf = len(fields)
         ny, nx = self.ny_local, self.nx_local
         fb = self._fb
-        if "col_sbuf0" not in fb or fb["col_sbuf0"].numel() < 3 * ny:
-            for k in ("col_sbuf0", "col_rbuf0", "col_sbuf1", "col_rbuf1"):
-                fb[k] = torch.empty(3 * ny, dtype=self.dtype,
-                                    device=self.device)
-            fb["cor_sbuf"] = torch.empty(12, dtype=self.dtype,
-                                         device=self.device)
-            fb["cor_rbuf"] = torch.empty_like(fb["cor_sbuf"])
-        (wrap_sides, col_ops, row_ops, cor_ops,
-         cor_mask) = halo_exchange_schedule(g, nx, ny)
+        wrap_sides, col_ops, row_ops, cor_ops, cor_mask = cache["sched"]
 
         for side in wrap_sides:
             ext.halo_wrap(fields, side)
@@ -315,7 +372,7 @@ class ShallowWater:
 
         if not (col_ops or row_ops or cor_ops):
             return
-        comm_id = self.comm.rccl_handle()
+        comm_id = cache["comm_id"]
         ext.group_start()
         for k, st, rf, _, _ in col_ops:
             if st is not None:
