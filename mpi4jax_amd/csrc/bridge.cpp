@@ -612,6 +612,38 @@ void unpack_corners(std::vector<at::Tensor> fields, at::Tensor buf,
                         (int)mask, is_double ? 1 : 0, cur_stream());
 }
 
+// direct access to the merged staging kernels (unit tests pin them
+// against the single-purpose pack/unpack kernels above)
+void pack_halo(std::vector<at::Tensor> fields, int64_t wrap_side,
+               c10::optional<at::Tensor> cb0, int64_t c0,
+               c10::optional<at::Tensor> cb1, int64_t c1,
+               c10::optional<at::Tensor> cor) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  launch_pack_halo(ptrs, (int)fields.size(), ny, nx, (int)wrap_side,
+                   cb0 ? cb0->data_ptr() : nullptr, c0,
+                   cb1 ? cb1->data_ptr() : nullptr, c1,
+                   cor ? cor->data_ptr() : nullptr, is_double ? 1 : 0,
+                   cur_stream());
+}
+
+void unpack_halo(std::vector<at::Tensor> fields,
+                 c10::optional<at::Tensor> cb0, int64_t c0,
+                 c10::optional<at::Tensor> cb1, int64_t c1,
+                 c10::optional<at::Tensor> cor, int64_t cor_mask) {
+  void* ptrs[3];
+  bool is_double;
+  int64_t ny, nx;
+  collect_field_ptrs(fields, ptrs, is_double, ny, nx);
+  launch_unpack_halo(ptrs, (int)fields.size(), ny, nx,
+                     cb0 ? cb0->data_ptr() : nullptr, c0,
+                     cb1 ? cb1->data_ptr() : nullptr, c1,
+                     cor ? cor->data_ptr() : nullptr, (int)cor_mask,
+                     is_double ? 1 : 0, cur_stream());
+}
+
 // One-call halo-exchange executor for the fused shallow-water step.
 // Executes the schedule computed by parallel/grid.halo_exchange_schedule
 // on the host side once per model (models/shallow_water.py caches it);
@@ -642,27 +674,37 @@ void sw_exchange(std::vector<at::Tensor> fields,
               "bad sw_exchange schedule");
   hipStream_t stream = cur_stream();
 
-  // kernel-side periodic wraps (never remote)
+  // stage everything outbound in ONE launch: periodic wrap + up to two
+  // column packs + the corner pack (launch count bounds strong scaling)
+  int wrap_side = -1;
   for (int64_t side : wrap_sides) {
-    launch_halo_wrap(ptrs, nf, ny, nx, (int)side, is_double ? 1 : 0,
-                     stream);
+    wrap_side = wrap_side < 0 ? (int)side : 2;  // two entries = both sides
   }
-  // pack sends
+  void* pack_cb[2] = {nullptr, nullptr};
+  int64_t pack_c[2] = {0, 0};
   for (size_t i = 0; i < col_ops.size(); i += 5) {
     if (col_ops[i + 1] < 0) continue;  // no send peer
-    const at::Tensor& sb = col_bufs.at(2 * col_ops[i]);
+    int64_t k = col_ops[i];
+    TORCH_CHECK(0 <= k && k < 2, "bad column buffer index");
+    const at::Tensor& sb = col_bufs.at(2 * k);
     TORCH_CHECK(sb.is_cuda() && sb.is_contiguous() &&
-                    sb.numel() >= (int64_t)nf * ny,
+                    sb.numel() >= (int64_t)nf * ny &&
+                    sb.scalar_type() == fields[0].scalar_type(),
                 "bad column send buffer");
-    launch_pack_cols(sb.data_ptr(), ptrs, nf, ny, nx, col_ops[i + 3],
-                     is_double ? 1 : 0, stream);
+    pack_cb[k] = sb.data_ptr();
+    pack_c[k] = col_ops[i + 3];
   }
+  void* cor_pack = nullptr;
   if (!cor_ops.empty()) {
     TORCH_CHECK(cor_sbuf.is_cuda() && cor_sbuf.is_contiguous() &&
                     cor_sbuf.numel() >= 4 * nf,
                 "bad corner send buffer");
-    launch_pack_corners(cor_sbuf.data_ptr(), ptrs, nf, ny, nx,
-                        is_double ? 1 : 0, stream);
+    cor_pack = cor_sbuf.data_ptr();
+  }
+  if (wrap_side >= 0 || pack_cb[0] || pack_cb[1] || cor_pack) {
+    launch_pack_halo(ptrs, nf, ny, nx, wrap_side, pack_cb[0], pack_c[0],
+                     pack_cb[1], pack_c[1], cor_pack, is_double ? 1 : 0,
+                     stream);
   }
   bool any_remote = false;
   for (size_t i = 0; i < col_ops.size(); i += 5)
@@ -727,17 +769,21 @@ void sw_exchange(std::vector<at::Tensor> fields,
   }
   RCCL_CHECK(ncclGroupEnd());
 
-  // unpack receives (column writes land before corner writes — the one
-  // ordering requirement of halo_plan; stream order provides it)
+  // unpack receives in ONE launch ("corners win" over column corner
+  // cells — the kernel skips those column cells when a corner owns them)
+  void* un_cb[2] = {nullptr, nullptr};
+  int64_t un_c[2] = {0, 0};
   for (size_t i = 0; i < col_ops.size(); i += 5) {
     if (col_ops[i + 2] < 0) continue;
-    launch_unpack_cols(ptrs, col_bufs.at(2 * col_ops[i] + 1).data_ptr(),
-                       nf, ny, nx, col_ops[i + 4], is_double ? 1 : 0,
-                       stream);
+    int64_t k = col_ops[i];
+    TORCH_CHECK(0 <= k && k < 2, "bad column buffer index");
+    un_cb[k] = col_bufs.at(2 * k + 1).data_ptr();
+    un_c[k] = col_ops[i + 4];
   }
-  if (cor_mask) {
-    launch_unpack_corners(ptrs, cor_rbuf.data_ptr(), nf, ny, nx,
-                          (int)cor_mask, is_double ? 1 : 0, stream);
+  if (un_cb[0] || un_cb[1] || cor_mask) {
+    launch_unpack_halo(ptrs, nf, ny, nx, un_cb[0], un_c[0], un_cb[1],
+                       un_c[1], cor_mask ? cor_rbuf.data_ptr() : nullptr,
+                       (int)cor_mask, is_double ? 1 : 0, stream);
   }
 }
 
@@ -789,4 +835,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_corners", &pack_corners);
   m.def("unpack_corners", &unpack_corners);
   m.def("sw_exchange", &sw_exchange);
+  m.def("pack_halo", &pack_halo);
+  m.def("unpack_halo", &unpack_halo);
 }

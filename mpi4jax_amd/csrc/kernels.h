@@ -74,3 +74,19 @@ void launch_pack_corners(void* buf, void* const* fields, int nf,
 void launch_unpack_corners(void* const* fields, const void* buf, int nf,
                            long long ny, long long nx, int mask,
                            int is_double, hipStream_t stream);
+
+// merged halo staging: one launch covering the periodic wrap, both column
+// packs and the corner pack (resp. the column + corner unpacks) — the
+// per-exchange launch count is what bounds strong scaling once the local
+// domain is small.  Null buffer => that segment is skipped; wrap_side -1
+// => no wrap.  The unpack skips a column's j=0 / j=ny-1 cell when the
+// corner segment of the SAME launch writes it (cor_mask bit set), which
+// preserves the "corners win" ordering rule of parallel/grid.halo_plan.
+void launch_pack_halo(void* const* fields, int nf, long long ny,
+                      long long nx, int wrap_side, void* cb0, long long c0,
+                      void* cb1, long long c1, void* cor, int is_double,
+                      hipStream_t stream);
+void launch_unpack_halo(void* const* fields, int nf, long long ny,
+                        long long nx, void* cb0, long long c0, void* cb1,
+                        long long c1, void* cor, int cor_mask,
+                        int is_double, hipStream_t stream);

@@ -976,6 +976,115 @@ __global__ void unpack_corners_kernel(HaloArgs<T> a, const T* buf,
   halo_field(a, f)[(long long)j * nx + i] = buf[t];
 }
 
+// merged staging: wrap + both column packs + corner pack in ONE launch
+// (work layout [wrap | cb0 | cb1 | corners]); see kernels.h for the
+// contract.  All segments touch disjoint cells: wrap writes halo cols
+// 0/nx-1, col packs read interior cols (1/nx-2), corner pack reads
+// interior corner cells.
+template <typename T>
+struct HaloXArgs {
+  T* f0;
+  T* f1;
+  T* f2;
+  int nf, ny, nx;
+  T* cb0;
+  T* cb1;
+  T* cor;
+  int c0, c1;
+  int wrap_side;  // -1 none, 0 east<-1, 1 west<-nx-2, 2 both
+  int cor_mask;
+};
+
+template <typename T>
+__device__ inline T* hx_field(const HaloXArgs<T>& a, int f) {
+  return f == 0 ? a.f0 : (f == 1 ? a.f1 : a.f2);
+}
+
+template <typename T>
+__global__ void pack_halo_kernel(HaloXArgs<T> a) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  const int ncol = a.nf * a.ny;
+  if (t < ncol) {
+    if (a.wrap_side < 0) return;
+    int f = t / a.ny, j = t % a.ny;
+    T* p = hx_field(a, f) + (long long)j * a.nx;
+    if (a.wrap_side != 1) p[a.nx - 1] = p[1];
+    if (a.wrap_side != 0) p[0] = p[a.nx - 2];
+    return;
+  }
+  t -= ncol;
+  if (t < 2 * ncol) {
+    T* buf = t < ncol ? a.cb0 : a.cb1;
+    int col = t < ncol ? a.c0 : a.c1;
+    if (!buf) return;
+    int s = t % ncol;
+    int f = s / a.ny, j = s % a.ny;
+    buf[s] = hx_field(a, f)[(long long)j * a.nx + col];
+    return;
+  }
+  t -= 2 * ncol;
+  if (t >= 4 * a.nf || !a.cor) return;
+  int d = t / a.nf, f = t % a.nf;
+  int j = (d < 2) ? 1 : a.ny - 2;
+  int i = (d == 0 || d == 2) ? 1 : a.nx - 2;
+  a.cor[t] = hx_field(a, f)[(long long)j * a.nx + i];
+}
+
+// which corner diagonal writes cell (j in {0, ny-1}, col in {0, nx-1}):
+// d0 (ny-1,nx-1), d1 (ny-1,0), d2 (0,nx-1), d3 (0,0)
+__device__ inline int corner_diag(bool top_row, bool east_col) {
+  return top_row ? (east_col ? 0 : 1) : (east_col ? 2 : 3);
+}
+
+template <typename T>
+__global__ void unpack_halo_kernel(HaloXArgs<T> a) {
+  int t = blockIdx.x * blockDim.x + threadIdx.x;
+  const int ncol = a.nf * a.ny;
+  if (t < 2 * ncol) {
+    const T* buf = t < ncol ? a.cb0 : a.cb1;
+    int col = t < ncol ? a.c0 : a.c1;
+    if (!buf) return;
+    int s = t % ncol;
+    int f = s / a.ny, j = s % a.ny;
+    if ((j == 0 || j == a.ny - 1) && a.cor) {
+      // the corner segment of this same launch owns this cell
+      int d = corner_diag(j == a.ny - 1, col == a.nx - 1);
+      if (a.cor_mask & (1 << d)) return;
+    }
+    hx_field(a, f)[(long long)j * a.nx + col] = buf[s];
+    return;
+  }
+  t -= 2 * ncol;
+  if (t >= 4 * a.nf || !a.cor) return;
+  int d = t / a.nf, f = t % a.nf;
+  if (!(a.cor_mask & (1 << d))) return;
+  int j = (d < 2) ? a.ny - 1 : 0;
+  int i = (d == 0 || d == 2) ? a.nx - 1 : 0;
+  hx_field(a, f)[(long long)j * a.nx + i] = a.cor[t];
+}
+
+template <typename T>
+HaloXArgs<T> make_hx_args(void* const* fields, int nf, long long ny,
+                          long long nx, int wrap_side, void* cb0,
+                          long long c0, void* cb1, long long c1, void* cor,
+                          int cor_mask) {
+  HaloXArgs<T> a;
+  a.f0 = (T*)fields[0];
+  a.f1 = nf > 1 ? (T*)fields[1] : (T*)fields[0];
+  a.f2 = nf > 2 ? (T*)fields[2] : (T*)fields[0];
+  a.nf = nf;
+  a.ny = (int)ny;
+  a.nx = (int)nx;
+  a.cb0 = (T*)cb0;
+  a.cb1 = (T*)cb1;
+  a.cor = (T*)cor;
+  a.c0 = (int)c0;
+  a.c1 = (int)c1;
+  a.wrap_side = wrap_side;
+  a.cor_mask = cor_mask;
+  return a;
+}
+
 }  // namespace
 
 void launch_pack_corners(void* buf, void* const* fields, int nf,
@@ -1005,5 +1114,41 @@ void launch_unpack_corners(void* const* fields, const void* buf, int nf,
     auto a = make_halo_args<float>(fields, nf, ny, nx, 0);
     hipLaunchKernelGGL(unpack_corners_kernel<float>, dim3(1), dim3(64), 0,
                        stream, a, (const float*)buf, mask);
+  }
+}
+
+void launch_pack_halo(void* const* fields, int nf, long long ny,
+                      long long nx, int wrap_side, void* cb0, long long c0,
+                      void* cb1, long long c1, void* cor, int is_double,
+                      hipStream_t stream) {
+  int n = 3 * nf * (int)ny + 4 * nf;
+  if (is_double) {
+    auto a = make_hx_args<double>(fields, nf, ny, nx, wrap_side, cb0, c0,
+                                  cb1, c1, cor, 0);
+    hipLaunchKernelGGL(pack_halo_kernel<double>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a);
+  } else {
+    auto a = make_hx_args<float>(fields, nf, ny, nx, wrap_side, cb0, c0,
+                                 cb1, c1, cor, 0);
+    hipLaunchKernelGGL(pack_halo_kernel<float>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a);
+  }
+}
+
+void launch_unpack_halo(void* const* fields, int nf, long long ny,
+                        long long nx, void* cb0, long long c0, void* cb1,
+                        long long c1, void* cor, int cor_mask,
+                        int is_double, hipStream_t stream) {
+  int n = 2 * nf * (int)ny + 4 * nf;
+  if (is_double) {
+    auto a = make_hx_args<double>(fields, nf, ny, nx, -1, cb0, c0, cb1, c1,
+                                  cor, cor_mask);
+    hipLaunchKernelGGL(unpack_halo_kernel<double>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a);
+  } else {
+    auto a = make_hx_args<float>(fields, nf, ny, nx, -1, cb0, c0, cb1, c1,
+                                 cor, cor_mask);
+    hipLaunchKernelGGL(unpack_halo_kernel<float>, dim3(halo_grid(n)),
+                       dim3(kBlock), 0, stream, a);
   }
 }

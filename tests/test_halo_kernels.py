@@ -190,3 +190,69 @@ def test_fused_model_python_executor_matches(monkeypatch):
     b = run(True)
     for f in ("h", "u", "v"):
         assert torch.equal(getattr(a, f), getattr(b, f)), f
+
+
+# ---------------------------------------------------------------------------
+# merged staging kernels (pack_halo / unpack_halo): one launch must equal
+# the single-purpose kernels run in sequence.
+
+@pytest.mark.parametrize("wrap", [-1, 0, 1, 2])
+@pytest.mark.parametrize("nf", [1, 3])
+def test_pack_halo_matches_separate(ext, wrap, nf):
+    fs = fields(nf=nf)
+    ref = [f.clone() for f in fs]
+    ny, nx = fs[0].shape
+    cb0 = torch.zeros(nf * ny, device="cuda")
+    cb1 = torch.zeros(nf * ny, device="cuda")
+    cor = torch.zeros(4 * nf, device="cuda")
+    ext.pack_halo(fs, wrap, cb0, 1, cb1, nx - 2, cor)
+
+    e_cb0 = torch.zeros_like(cb0)
+    e_cb1 = torch.zeros_like(cb1)
+    e_cor = torch.zeros_like(cor)
+    if wrap >= 0:
+        ext.halo_wrap(ref, wrap)
+    ext.pack_cols(e_cb0, ref, 1)
+    ext.pack_cols(e_cb1, ref, nx - 2)
+    ext.pack_corners(e_cor, ref)
+    torch.cuda.synchronize()
+    assert torch.equal(cb0, e_cb0)
+    assert torch.equal(cb1, e_cb1)
+    assert torch.equal(cor, e_cor)
+    for f, r in zip(fs, ref):
+        assert torch.equal(f, r)  # wrap applied identically
+
+
+def test_pack_halo_skips_null_segments(ext):
+    fs = fields()
+    before = [f.clone() for f in fs]
+    ny = fs[0].shape[0]
+    cb1 = torch.full((3 * ny,), -7.0, device="cuda")
+    ext.pack_halo(fs, -1, None, 0, cb1, 1, None)
+    torch.cuda.synchronize()
+    for f, b in zip(fs, before):
+        assert torch.equal(f, b)
+    assert not torch.equal(cb1, torch.full_like(cb1, -7.0))
+
+
+@pytest.mark.parametrize("cor_mask", [0, 0b1111, 0b0101])
+def test_unpack_halo_matches_separate(ext, cor_mask):
+    nf = 3
+    fs = fields(nf=nf)
+    ref = [f.clone() for f in fs]
+    ny, nx = fs[0].shape
+    g = torch.Generator().manual_seed(7)
+    cb0 = torch.randn(nf * ny, generator=g).cuda()
+    cb1 = torch.randn(nf * ny, generator=g).cuda()
+    cor = torch.randn(4 * nf, generator=g).cuda()
+    ext.unpack_halo(fs, cb0, nx - 1, cb1, 0, cor if cor_mask else None,
+                    cor_mask)
+
+    # reference order: columns land first, corners overwrite
+    ext.unpack_cols(ref, cb0, nx - 1)
+    ext.unpack_cols(ref, cb1, 0)
+    if cor_mask:
+        ext.unpack_corners(ref, cor, cor_mask)
+    torch.cuda.synchronize()
+    for f, r in zip(fs, ref):
+        assert torch.equal(f, r)
