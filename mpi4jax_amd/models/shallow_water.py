@@ -403,38 +403,64 @@ class ShallowWater:
         """Return ``(advance, current)``: ``advance()`` runs
         ``steps_per_call`` model steps and returns the current state.
 
-        On the fused GPU path the steps are captured into a single hipGraph
-        (collectives included — RCCL is capture-safe), so a whole multistep
-        replays with one launch from the host.  ``steps_per_call`` must be
-        even (buffer parity).  Falls back to an eager loop if capture is
-        unavailable.
+        On the fused GPU path the steps are captured into a single
+        hipGraph (halo-exchange RCCL enqueues included — validated by
+        tools/probe_rccl_graph.py), so a whole multistep replays with one
+        launch from the host.  ``steps_per_call`` must be even (buffer
+        parity).  At any world size the capture is adopted only after a
+        replay reproduces the eager trajectory bitwise, with all-or-none
+        agreement across ranks; otherwise every rank falls back to the
+        (wire-identical) eager loop.  ``MPI4JAX_AMD_SW_GRAPH=0`` forces
+        the eager loop.
         """
+        import os
+
+        env = os.environ.get("MPI4JAX_AMD_SW_GRAPH", "").strip()
         if use_graph is None:
-            # graphs only at world size 1 for now: RCCL capture works in
-            # principle, but a capture that succeeds yet replays a stale
-            # p2p schedule would corrupt halos silently — keep multi-rank
-            # on the plain fused loop until graph replay is validated on a
-            # multi-GPU box
             use_graph = (self.fused and self.device.type == "cuda"
-                         and self.comm.size == 1)
+                         and env != "0")
         if use_graph and steps_per_call % 2:
             raise ValueError("steps_per_call must be even for graph capture")
 
-        if not use_graph:
-            holder = {"s": state}
+        def eager_stepper(s):
+            holder = {"s": s}
 
             def advance():
                 for _ in range(steps_per_call):
                     holder["s"] = self.step(holder["s"])
                 return holder["s"]
 
-            return advance, state
+            return advance, s
+
+        if not use_graph:
+            return eager_stepper(state)
 
         # make sure the fused buffers exist & are warm (2 steps, also
-        # pre-allocating every temp the exchange path uses)
+        # pre-allocating every temp the exchange path uses and
+        # establishing every RCCL p2p connection the capture will record)
         state = self.step(state)
         state = self.step(state)
         torch.cuda.synchronize()
+
+        # Validate-then-adopt protocol (all-or-none across ranks): record
+        # the eager trajectory, rewind, capture the same steps into a
+        # hipGraph, replay once and require a bitwise match.  Both
+        # agreement points are plain allreduces so no rank ever replays a
+        # graph while another runs the eager loop with a half-captured
+        # schedule; on any failure every rank rewinds and uses the eager
+        # loop (wire-identical per step, so even that path stays matched).
+        fb = self._fb
+        keys = ("h", "u", "v", "do_h", "do_u", "do_v")
+        s0 = {k: fb[k].clone() for k in keys}
+        s = state
+        for _ in range(steps_per_call):
+            s = self.step(s)
+        torch.cuda.synchronize()
+        ref = {k: fb[k].clone() for k in keys}
+        for k in keys:
+            fb[k].copy_(s0[k])
+        torch.cuda.synchronize()
+
         try:
             graph = torch.cuda.CUDAGraph()
             with torch.cuda.graph(graph):
@@ -442,24 +468,44 @@ class ShallowWater:
                 for _ in range(steps_per_call):
                     s = self.step(s)
             final = s
+            captured = True
         except Exception:
-            # capture unsupported (e.g. RCCL version without graph
-            # support) — eager fallback
+            captured = False
             torch.cuda.synchronize()
-            holder = {"s": state}
 
-            def advance():
-                for _ in range(steps_per_call):
-                    holder["s"] = self.step(holder["s"])
-                return holder["s"]
+        if not self._all_ranks_agree(captured):
+            for k in keys:
+                fb[k].copy_(s0[k])
+            torch.cuda.synchronize()
+            return eager_stepper(state)
 
-            return advance, state
+        graph.replay()
+        torch.cuda.synchronize()
+        matches = all(torch.equal(fb[k], ref[k]) for k in keys)
+        # rewind so the net effect of make_stepper is always exactly the
+        # two warm-up steps, whichever path was adopted
+        for k in keys:
+            fb[k].copy_(s0[k])
+        torch.cuda.synchronize()
+        if not self._all_ranks_agree(matches):
+            return eager_stepper(state)
 
         def advance():
             graph.replay()
             return final
 
         return advance, state
+
+    def _all_ranks_agree(self, ok):
+        """MIN-allreduce of a local predicate over the model's comm."""
+        if self.comm.size == 1:
+            return bool(ok)
+        from ..ops.allreduce import allreduce
+        from ..ops.reduce_ops import MIN
+
+        flag = torch.tensor([1 if ok else 0], dtype=torch.int32,
+                            device=self.device)
+        return bool(allreduce(flag, MIN, comm=self.comm).item() == 1)
 
     def _step_eager(self, state, first_step=False):
         h, u, v, dh, du, dv = state

@@ -327,6 +327,59 @@ def test_graph_stepper_matches_plain_fused():
         assert torch.equal(a, b), (name, (a - b).abs().max().item())
 
 
+def test_graph_env_disable_matches(monkeypatch):
+    """MPI4JAX_AMD_SW_GRAPH=0 must run the eager loop with identical
+    results (the stepper contract is independent of the path taken)."""
+    from mpi4jax_amd.models import ShallowWater
+
+    def run():
+        sw = ShallowWater(nx=96, ny=48, device="cuda",
+                          comm=m.get_world().Clone())
+        s = sw.initial_conditions()
+        s = sw.step(s, first_step=True)
+        advance, s = sw.make_stepper(s, steps_per_call=2)
+        s = advance()
+        s = advance()
+        torch.cuda.synchronize()
+        return s
+
+    a = run()
+    monkeypatch.setenv("MPI4JAX_AMD_SW_GRAPH", "0")
+    b = run()
+    for name in ("h", "u", "v"):
+        assert torch.equal(getattr(a, name), getattr(b, name)), name
+
+
+def test_graph_capture_failure_falls_back(monkeypatch):
+    """If hipGraph capture raises, make_stepper must rewind and adopt the
+    eager loop — trajectory identical to a never-captured run."""
+    from mpi4jax_amd.models import ShallowWater
+
+    def run(break_capture):
+        if break_capture:
+            class _Boom:
+                def __init__(self, *a, **k):
+                    raise RuntimeError("capture unavailable (test)")
+
+            monkeypatch.setattr(torch.cuda, "CUDAGraph", _Boom)
+        else:
+            monkeypatch.undo()
+        sw = ShallowWater(nx=96, ny=48, device="cuda",
+                          comm=m.get_world().Clone())
+        s = sw.initial_conditions()
+        s = sw.step(s, first_step=True)
+        advance, s = sw.make_stepper(s, steps_per_call=2)
+        s = advance()
+        s = advance()
+        torch.cuda.synchronize()
+        return s
+
+    a = run(False)
+    b = run(True)
+    for name in ("h", "u", "v"):
+        assert torch.equal(getattr(a, name), getattr(b, name)), name
+
+
 def test_allreduce_bandwidth_smoke():
     """256 MiB bf16 in-stream allreduce completes and is identity at n=1."""
     n = 128 * 1024 * 1024  # 256 MiB of bf16
