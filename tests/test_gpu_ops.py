@@ -332,20 +332,24 @@ def test_graph_env_disable_matches(monkeypatch):
     results (the stepper contract is independent of the path taken)."""
     from mpi4jax_amd.models import ShallowWater
 
-    def run():
+    def run(extra_warm):
         sw = ShallowWater(nx=96, ny=48, device="cuda",
                           comm=m.get_world().Clone())
         s = sw.initial_conditions()
         s = sw.step(s, first_step=True)
+        # the graph path consumes two warm-up steps inside make_stepper;
+        # match the totals when it is disabled
+        for _ in range(extra_warm):
+            s = sw.step(s)
         advance, s = sw.make_stepper(s, steps_per_call=2)
         s = advance()
         s = advance()
         torch.cuda.synchronize()
         return s
 
-    a = run()
+    a = run(0)
     monkeypatch.setenv("MPI4JAX_AMD_SW_GRAPH", "0")
-    b = run()
+    b = run(2)
     for name in ("h", "u", "v"):
         assert torch.equal(getattr(a, name), getattr(b, name)), name
 
