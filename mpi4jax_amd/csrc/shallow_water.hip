@@ -385,6 +385,14 @@ typedef float vf2 __attribute__((ext_vector_type(2), aligned(4)));
 __device__ inline vf2 ld2(const float* p, long long off) {
   return *(const vf2*)(p + off);
 }
+__device__ inline void st2(float* p, long long off, vf2 v) {
+  *(vf2*)(p + off) = v;
+}
+// float2 lane extracts from a float4 loaded at idx-1 (lane k = offset
+// k-1): used by the 2-column stage variant
+__device__ inline vf2 x01(vf4 A) { return (vf2){A.x, A.y}; }
+__device__ inline vf2 x12(vf4 A) { return (vf2){A.y, A.z}; }
+__device__ inline vf2 x23(vf4 A) { return (vf2){A.z, A.w}; }
 
 // shifted-vector builders: lane c of the result holds value at i0+c+k
 __device__ inline vf4 sh0(vf4 Am1, float x3) {
@@ -407,6 +415,9 @@ __device__ inline vf4 rcp4(vf4 x) {
   // chains (~30 unpipelined cycles each) from the per-pack critical path
   return (vf4){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y),
                __builtin_amdgcn_rcpf(x.z), __builtin_amdgcn_rcpf(x.w)};
+}
+__device__ inline vf2 rcp2(vf2 x) {
+  return (vf2){__builtin_amdgcn_rcpf(x.x), __builtin_amdgcn_rcpf(x.y)};
 }
 
 
@@ -750,6 +761,103 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   st4(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
 }
 
+// stage 19 = stage 18 at 2 columns/thread (float2 math, float4 row
+// loads).  stage18v holds 118 VGPRs -> 4 waves/SIMD and parks ~48% of
+// cycles on memory (profiles/README.md); halving the per-thread state
+// trades ILP for occupancy to hide that latency.  Same operation order
+// as stage18v, so the trajectories agree to float rounding.
+__global__ void sw_stage19v(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
+  if (t >= ppr * ny) return;
+  const int j = t / ppr;
+  const int i0 = (t % ppr) * 2;
+  const int jmin = a.f.south_open ? 1 : 2;
+  const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
+  const int imin = a.f.west_open ? 1 : 2;
+  const int imax = (a.f.east_open && !a.f.east_wall) ? nx - 1 : nx - 2;
+  // the ld4 at idn reads offsets 0..3; keep it in-row (i0+3 <= nx-1)
+  const bool fast = j >= jmin && j <= jmax && i0 >= imin &&
+                    i0 + 2 <= imax && i0 + 3 < nx;
+  if (!fast) {
+    for (int c = 0; c < 2 && i0 + c < nx; ++c) stage8_cell(a, j, i0 + c);
+    return;
+  }
+  const long long idx = (long long)j * nx + i0;
+  const long long idn = idx - nx, idp = idx + nx;
+
+  // all loads first (one vmcnt window), float4 per row
+  vf4 HmA = ld4(a.h, idn);      // h[j-1][0..3]
+  vf4 H0A = ld4(a.h, idx - 1);  // h[j][-1..2]
+  vf4 HpA = ld4(a.h, idp - 1);  // h[j+1][-1..2]
+  vf2 Um0 = ld2(a.u, idn);      // u[j-1][0..1]
+  vf4 U0A = ld4(a.u, idx - 1);  // u[j][-1..2]
+  vf4 UpA = ld4(a.u, idp - 1);  // u[j+1][-1..2]
+  vf4 VmA = ld4(a.v, idn);      // v[j-1][0..3]
+  vf4 V0A = ld4(a.v, idx - 1);  // v[j][-1..2]
+  vf2 Vp0 = ld2(a.v, idp);      // v[j+1][0..1]
+  vf2 doh = ld2(a.doh, idx), dou = ld2(a.dou, idx), dov = ld2(a.dov, idx);
+
+  vf2 Hm0 = x01(HmA), Hm1 = x12(HmA);
+  vf2 H0m1 = x01(H0A), H00 = x12(H0A), H01 = x23(H0A);
+  vf2 Hpm1 = x01(HpA), Hp0 = x12(HpA), Hp1 = x23(HpA);
+  vf2 U0m1 = x01(U0A), U00 = x12(U0A), U01 = x23(U0A);
+  vf2 Upm1 = x01(UpA), Up0 = x12(UpA);
+  vf2 Vm0 = x01(VmA), Vm1 = x12(VmA);
+  vf2 V0m1 = x01(V0A), V00 = x12(V0A), V01 = x23(V0A);
+
+  const float rdx = a.rdx, rdy = a.rdy;
+
+  vf2 fe_c = 0.5f * (H00 + H01) * U00;
+  vf2 fe_w = 0.5f * (H0m1 + H00) * U0m1;
+  vf2 fe_n = 0.5f * (Hp0 + Hp1) * Up0;
+  vf2 fe_nw = 0.5f * (Hpm1 + Hp0) * Upm1;
+  vf2 fn_c = 0.5f * (H00 + Hp0) * V00;
+  vf2 fn_e = 0.5f * (H01 + Hp1) * V01;
+  vf2 fn_s = 0.5f * (Hm0 + H00) * Vm0;
+  vf2 fn_se = 0.5f * (Hm1 + H01) * Vm1;
+
+  float corj = a.cor_base + (float)j * a.cor_dj;
+  float corjm = a.cor_base + (float)(j - 1) * a.cor_dj;
+  vf2 q_c = corj + ((V01 - V00) * rdx - (Up0 - U00) * rdy);
+  q_c *= rcp2(0.25f * (H00 + H01 + Hp0 + Hp1));
+  vf2 q_s = corjm + ((Vm1 - Vm0) * rdx - (U00 - Um0) * rdy);
+  q_s *= rcp2(0.25f * (Hm0 + Hm1 + H00 + H01));
+  vf2 q_w = corj + ((V00 - V0m1) * rdx - (Upm1 - U0m1) * rdy);
+  q_w *= rcp2(0.25f * (H0m1 + H00 + Hpm1 + Hp0));
+
+  vf2 ke_c = 0.5f * (0.5f * (U00 * U00 + U0m1 * U0m1) +
+                     0.5f * (V00 * V00 + Vm0 * Vm0));
+  vf2 ke_e = 0.5f * (0.5f * (U01 * U01 + U00 * U00) +
+                     0.5f * (V01 * V01 + Vm1 * Vm1));
+  vf2 ke_n = 0.5f * (0.5f * (Up0 * Up0 + Upm1 * Upm1) +
+                     0.5f * (Vp0 * Vp0 + V00 * V00));
+
+  vf2 dnh = -(fe_c - fe_w) * rdx - (fn_c - fn_s) * rdy;
+  vf2 dnu = -G * (H01 - H00) * rdx +
+            0.5f * (q_c * 0.5f * (fn_c + fn_e) +
+                    q_s * 0.5f * (fn_s + fn_se));
+  dnu -= (ke_e - ke_c) * rdx;
+  vf2 dnv = -G * (Hp0 - H00) * rdy -
+            0.5f * (q_c * 0.5f * (fe_c + fe_n) +
+                    q_w * 0.5f * (fe_w + fe_nw));
+  dnv -= (ke_n - ke_c) * rdy;
+
+  st2(a.dnh, idx, dnh);
+  st2(a.dnu, idx, dnu);
+  st2(a.dnv, idx, dnv);
+  st2(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
+  st2(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
+  st2(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
+}
+
 int sw_grid(long long n) {
   long long blocks = (n + kBlock - 1) / kBlock;
   if (blocks > 4096) blocks = 4096;  // grid-stride beyond this
@@ -817,7 +925,7 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
     // vectorized float stages: 11 -> stage1v, 16 -> stage6v, 17 -> stage7v
     SwArgs<float> a;
     sw_fill_args(a, p);
-    long long ppr = (p.nx + 3) / 4;
+    long long ppr = stage == 19 ? (p.nx + 1) / 2 : (p.nx + 3) / 4;
     long long packs = ppr * p.ny;
     dim3 grid((unsigned)((packs + 255) / 256)), block(256);
     switch (stage) {
@@ -825,6 +933,7 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 16: hipLaunchKernelGGL(sw_stage6v, grid, block, 0, stream, a); break;
       case 17: hipLaunchKernelGGL(sw_stage7v, grid, block, 0, stream, a); break;
       case 18: hipLaunchKernelGGL(sw_stage18v, grid, block, 0, stream, a); break;
+      case 19: hipLaunchKernelGGL(sw_stage19v, grid, block, 0, stream, a); break;
     }
     return;
   }

@@ -241,6 +241,8 @@ class ShallowWater:
                 and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
             if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
                 return 11, 16, 17
+            if os.environ.get("MPI4JAX_AMD_SW_V2"):
+                return None, 19, 17  # 2-col variant (occupancy experiment)
             return None, 18, 17  # merged+vectorized single pass
         return 1, 6, 7
 
