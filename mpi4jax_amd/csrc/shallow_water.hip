@@ -761,6 +761,47 @@ __global__ void sw_stage18v(SwArgs<float> a) {
   st4(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
 }
 
+// stage 27 = stage 17 (friction Laplacian) at 2 columns/thread — same
+// occupancy rationale as stage 19 below.
+__global__ void sw_stage27v(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
+  if (t >= ppr * ny) return;
+  const int j = t / ppr;
+  const int i0 = (t % ppr) * 2;
+  const int jmin = a.f.south_open ? 1 : 2;
+  const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
+  const int imin = a.f.west_open ? 1 : 2;
+  const int imax = (a.f.east_open && !a.f.east_wall) ? nx - 1 : nx - 2;
+  const bool fast = j >= jmin && j <= jmax && i0 >= imin &&
+                    i0 + 2 <= imax && i0 + 2 < nx;
+  if (!fast) {
+    for (int c = 0; c < 2 && i0 + c < nx; ++c) stage7_cell(a, j, i0 + c);
+    return;
+  }
+  const long long idx = (long long)j * nx + i0;
+  vf4 U0A = ld4(a.u, idx - 1);
+  vf2 un = ld2(a.u, idx + nx), us = ld2(a.u, idx - nx);
+  vf4 V0A = ld4(a.v, idx - 1);
+  vf2 vn = ld2(a.v, idx + nx), vs = ld2(a.v, idx - nx);
+  vf2 uw = x01(U0A), uc = x12(U0A), ue = x23(U0A);
+  vf2 vw = x01(V0A), vcc = x12(V0A), ve = x23(V0A);
+  const float nu = a.nu;
+  vf2 lu = (nu * (ue - uc) * a.rdx - nu * (uc - uw) * a.rdx) * a.rdx +
+           (nu * (un - uc) * a.rdy - nu * (uc - us) * a.rdy) * a.rdy;
+  vf2 lv = (nu * (ve - vcc) * a.rdx - nu * (vcc - vw) * a.rdx) * a.rdx +
+           (nu * (vn - vcc) * a.rdy - nu * (vcc - vs) * a.rdy) * a.rdy;
+  st2(a.u2, idx, uc + a.dt * lu);
+  st2(a.v2, idx, vcc + a.dt * lv);
+}
+
 // stage 19 = stage 18 at 2 columns/thread (float2 math, float4 row
 // loads).  stage18v holds 118 VGPRs -> 4 waves/SIMD and parks ~48% of
 // cycles on memory (profiles/README.md); halving the per-thread state
@@ -925,7 +966,7 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
     // vectorized float stages: 11 -> stage1v, 16 -> stage6v, 17 -> stage7v
     SwArgs<float> a;
     sw_fill_args(a, p);
-    long long ppr = stage == 19 ? (p.nx + 1) / 2 : (p.nx + 3) / 4;
+    long long ppr = stage >= 19 ? (p.nx + 1) / 2 : (p.nx + 3) / 4;
     long long packs = ppr * p.ny;
     dim3 grid((unsigned)((packs + 255) / 256)), block(256);
     switch (stage) {
@@ -934,6 +975,7 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 17: hipLaunchKernelGGL(sw_stage7v, grid, block, 0, stream, a); break;
       case 18: hipLaunchKernelGGL(sw_stage18v, grid, block, 0, stream, a); break;
       case 19: hipLaunchKernelGGL(sw_stage19v, grid, block, 0, stream, a); break;
+      case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
     }
     return;
   }

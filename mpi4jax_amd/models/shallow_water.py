@@ -241,9 +241,11 @@ class ShallowWater:
                 and not os.environ.get("MPI4JAX_AMD_SW_NOVEC")):
             if os.environ.get("MPI4JAX_AMD_SW_TWOPASS"):
                 return 11, 16, 17
-            if os.environ.get("MPI4JAX_AMD_SW_V2"):
-                return None, 19, 17  # 2-col variant (occupancy experiment)
-            return None, 18, 17  # merged+vectorized single pass
+            if os.environ.get("MPI4JAX_AMD_SW_4COL"):
+                return None, 18, 17  # 4-col variants (4 waves/SIMD)
+            # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
+            # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
+            return None, 19, 27
         return 1, 6, 7
 
     def _step_fused(self, state, first_step=False):

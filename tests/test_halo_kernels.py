@@ -259,15 +259,15 @@ def test_unpack_halo_matches_separate(ext, cor_mask):
 
 
 def test_v2_stage_variant_matches(monkeypatch):
-    """The 2-column stage-19 kernel must reproduce the stage-18 default
-    (same operation order; float rounding only)."""
+    """The 2-column default kernels (stages 19/27) must reproduce the
+    4-column variants (same operation order; float rounding only)."""
     from mpi4jax_amd.models import ShallowWater
 
-    def run(v2):
-        if v2:
-            monkeypatch.setenv("MPI4JAX_AMD_SW_V2", "1")
+    def run(wide):
+        if wide:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_4COL", "1")
         else:
-            monkeypatch.delenv("MPI4JAX_AMD_SW_V2", raising=False)
+            monkeypatch.delenv("MPI4JAX_AMD_SW_4COL", raising=False)
         sw = ShallowWater(nx=130, ny=66, device="cuda")
         s = sw.initial_conditions()
         s = sw.step(s, first_step=True)
@@ -285,14 +285,15 @@ def test_v2_stage_variant_matches(monkeypatch):
 
 
 def test_v2_stage_variant_walls(monkeypatch):
-    """Stage-19 with closed east boundary (wall masks + scalar edge)."""
+    """2-col kernels with closed east boundary (wall masks + scalar
+    edge columns)."""
     from mpi4jax_amd.models import ShallowWater
 
-    def run(v2):
-        if v2:
-            monkeypatch.setenv("MPI4JAX_AMD_SW_V2", "1")
+    def run(wide):
+        if wide:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_4COL", "1")
         else:
-            monkeypatch.delenv("MPI4JAX_AMD_SW_V2", raising=False)
+            monkeypatch.delenv("MPI4JAX_AMD_SW_4COL", raising=False)
         sw = ShallowWater(nx=97, ny=49, device="cuda", periodic_x=False)
         s = sw.initial_conditions()
         s = sw.step(s, first_step=True)
