@@ -807,7 +807,31 @@ __global__ void sw_stage27v(SwArgs<float> a) {
 // cycles on memory (profiles/README.md); halving the per-thread state
 // trades ILP for occupancy to hide that latency.  Same operation order
 // as stage18v, so the trajectories agree to float rounding.
-__global__ void sw_stage19v(SwArgs<float> a) {
+//
+// NT variant (stage 20): the do* tendency loads and every store are
+// single-use streams consumed only on the NEXT step (by which point the
+// whole 312 MB working set has cycled through anyway), so nontemporal
+// hints keep them from evicting the h/u/v rows that DO have intra-step
+// reuse across j-neighboring threads.
+template <bool NT>
+__device__ inline vf2 ld2s(const float* p, long long off) {
+  if constexpr (NT) {
+    return __builtin_nontemporal_load((const vf2*)(p + off));
+  } else {
+    return *(const vf2*)(p + off);
+  }
+}
+template <bool NT>
+__device__ inline void st2s(float* p, long long off, vf2 v) {
+  if constexpr (NT) {
+    __builtin_nontemporal_store(v, (vf2*)(p + off));
+  } else {
+    *(vf2*)(p + off) = v;
+  }
+}
+
+template <bool NT>
+__global__ void sw_stage19t(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int ppr = (nx + 1) / 2;
   const int T_ = (int)gridDim.x;
@@ -844,7 +868,8 @@ __global__ void sw_stage19v(SwArgs<float> a) {
   vf4 VmA = ld4(a.v, idn);      // v[j-1][0..3]
   vf4 V0A = ld4(a.v, idx - 1);  // v[j][-1..2]
   vf2 Vp0 = ld2(a.v, idp);      // v[j+1][0..1]
-  vf2 doh = ld2(a.doh, idx), dou = ld2(a.dou, idx), dov = ld2(a.dov, idx);
+  vf2 doh = ld2s<NT>(a.doh, idx), dou = ld2s<NT>(a.dou, idx),
+      dov = ld2s<NT>(a.dov, idx);
 
   vf2 Hm0 = x01(HmA), Hm1 = x12(HmA);
   vf2 H0m1 = x01(H0A), H00 = x12(H0A), H01 = x23(H0A);
@@ -891,12 +916,12 @@ __global__ void sw_stage19v(SwArgs<float> a) {
                     q_w * 0.5f * (fe_w + fe_nw));
   dnv -= (ke_n - ke_c) * rdy;
 
-  st2(a.dnh, idx, dnh);
-  st2(a.dnu, idx, dnu);
-  st2(a.dnv, idx, dnv);
-  st2(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
-  st2(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
-  st2(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
+  st2s<NT>(a.dnh, idx, dnh);
+  st2s<NT>(a.dnu, idx, dnu);
+  st2s<NT>(a.dnv, idx, dnv);
+  st2s<NT>(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
+  st2s<NT>(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
+  st2s<NT>(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
 }
 
 int sw_grid(long long n) {
@@ -974,7 +999,8 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 16: hipLaunchKernelGGL(sw_stage6v, grid, block, 0, stream, a); break;
       case 17: hipLaunchKernelGGL(sw_stage7v, grid, block, 0, stream, a); break;
       case 18: hipLaunchKernelGGL(sw_stage18v, grid, block, 0, stream, a); break;
-      case 19: hipLaunchKernelGGL(sw_stage19v, grid, block, 0, stream, a); break;
+      case 19: hipLaunchKernelGGL(sw_stage19t<false>, grid, block, 0, stream, a); break;
+      case 20: hipLaunchKernelGGL(sw_stage19t<true>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
     }
     return;

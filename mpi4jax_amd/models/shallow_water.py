@@ -243,6 +243,8 @@ class ShallowWater:
                 return 11, 16, 17
             if os.environ.get("MPI4JAX_AMD_SW_4COL"):
                 return None, 18, 17  # 4-col variants (4 waves/SIMD)
+            if os.environ.get("MPI4JAX_AMD_SW_NT"):
+                return None, 20, 27  # + nontemporal streaming hints
             # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
             # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
             return None, 19, 27
