@@ -137,6 +137,9 @@ class _FakeComm:
     def __init__(self, rank, size):
         self.rank, self.size = rank, size
 
+    def rccl_handle(self):
+        return 1000 + self.rank  # placeholder; never dereferenced on CPU
+
 
 def test_halo_schedule_cross_rank_matching():
     """For every topology the driver can run (1..8 ranks), the fused
@@ -186,3 +189,80 @@ def test_halo_schedule_cross_rank_matching():
             tx = sends.get((src, dst), [])
             rx = recvs.get((dst, src), [])
             assert tx == rx, (dims, src, dst, tx, rx)
+
+
+def test_stage_plan_selection(monkeypatch):
+    """Env-var kernel-path selection resolved once at buffer init."""
+    import torch
+    from mpi4jax_amd.models import ShallowWater
+
+    sw = ShallowWater(nx=12, ny=6, device="cpu")
+    for var in ("MPI4JAX_AMD_SW_MERGED", "MPI4JAX_AMD_SW_NOVEC",
+                "MPI4JAX_AMD_SW_TWOPASS", "MPI4JAX_AMD_SW_4COL",
+                "MPI4JAX_AMD_SW_NT"):
+        monkeypatch.delenv(var, raising=False)
+    assert sw._stage_plan() == (None, 19, 27)
+    monkeypatch.setenv("MPI4JAX_AMD_SW_4COL", "1")
+    assert sw._stage_plan() == (None, 18, 17)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_4COL")
+    monkeypatch.setenv("MPI4JAX_AMD_SW_NT", "1")
+    assert sw._stage_plan() == (None, 20, 27)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_NT")
+    monkeypatch.setenv("MPI4JAX_AMD_SW_TWOPASS", "1")
+    assert sw._stage_plan() == (11, 16, 17)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_TWOPASS")
+    monkeypatch.setenv("MPI4JAX_AMD_SW_NOVEC", "1")
+    assert sw._stage_plan() == (1, 6, 7)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_NOVEC")
+    monkeypatch.setenv("MPI4JAX_AMD_SW_MERGED", "1")
+    assert sw._stage_plan() == (None, 8, 7)
+    # f64 always takes the scalar kernels
+    sw64 = ShallowWater(nx=12, ny=6, device="cpu", dtype=torch.float64)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_MERGED")
+    assert sw64._stage_plan() == (1, 6, 7)
+
+
+def test_exchange_cache_flat_encoding():
+    """The flattened int schedule handed to the native sw_exchange must
+    encode exactly the Python schedule with None -> -1, for every
+    topology (the C++ side only executes this encoding)."""
+    from mpi4jax_amd.models.shallow_water import ShallowWater
+    from mpi4jax_amd.parallel.grid import halo_exchange_schedule
+
+    ny, nx = 8, 10
+    for dims in [(1, 1), (2, 1), (2, 2), (2, 4), (3, 2), (1, 8)]:
+        size = dims[0] * dims[1]
+        for rank in range(size):
+            sw = ShallowWater.__new__(ShallowWater)
+            import torch as _t
+
+            from mpi4jax_amd.parallel.grid import CartesianGrid
+            g = CartesianGrid.__new__(CartesianGrid)
+            g.comm = _FakeComm(rank, size)
+            g.nproc_y, g.nproc_x = dims
+            g.periodic_y, g.periodic_x = False, True
+            g.coords = (rank // dims[1], rank % dims[1])
+            sw.grid = g
+            sw.comm = g.comm
+            sw.ny_local, sw.nx_local = ny, nx
+            sw.device = _t.device("cpu")
+            sw.dtype = _t.float32
+            sw._fb = {}
+            cache = sw._exchange_cache()
+            wrap_s, col_s, row_s, cor_s, mask_s = cache["sched"]
+            wrap_f, col_f, row_f, cor_f, mask_f = cache["flat"]
+            assert wrap_f == list(wrap_s) and mask_f == mask_s
+
+            def flatten(ops):
+                return [-1 if v is None else int(v)
+                        for op in ops for v in op]
+
+            assert col_f == flatten(col_s)
+            assert row_f == flatten(row_s)
+            assert cor_f == flatten(cor_s)
+            assert len(col_f) % 5 == 0
+            assert len(row_f) % 4 == 0
+            assert len(cor_f) % 3 == 0
+            # no remote peers at world 1 => comm never resolved
+            if size == 1:
+                assert cache["comm_id"] == -1
