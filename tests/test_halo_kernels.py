@@ -308,3 +308,27 @@ def test_v2_stage_variant_walls(monkeypatch):
         x, y = getattr(a, f), getattr(b, f)
         assert torch.allclose(x, y, atol=1e-5, rtol=1e-6), (
             f, (x - y).abs().max().item())
+
+
+def test_nt_stage_variant_matches(monkeypatch):
+    """The nontemporal-hint kernel (stage 20) is a pure cache-policy
+    change: results must be bitwise equal to the default stage 19."""
+    from mpi4jax_amd.models import ShallowWater
+
+    def run(nt):
+        if nt:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_NT", "1")
+        else:
+            monkeypatch.delenv("MPI4JAX_AMD_SW_NT", raising=False)
+        sw = ShallowWater(nx=130, ny=66, device="cuda")
+        s = sw.initial_conditions()
+        s = sw.step(s, first_step=True)
+        for _ in range(6):
+            s = sw.step(s)
+        torch.cuda.synchronize()
+        return s
+
+    a = run(False)
+    b = run(True)
+    for f in ("h", "u", "v"):
+        assert torch.equal(getattr(a, f), getattr(b, f)), f
