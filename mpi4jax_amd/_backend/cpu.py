@@ -30,6 +30,17 @@ def _wire(t):
         t.contiguous().flatten().view(torch.uint8)
 
 
+_BITWISE_OPS = (Op.BAND, Op.BOR, Op.BXOR)
+
+
+def _check_bitwise_dtype(op, x, op_name):
+    if op in _BITWISE_OPS and (x.is_floating_point() or x.is_complex()):
+        raise ValueError(
+            f"{op_name}: {op} requires an integer or bool dtype, "
+            f"got {x.dtype}"
+        )
+
+
 def _gloo_reduce_op(op: Op, dtype):
     if op is Op.AVG:
         # gloo has no AVG; emulate with SUM + divide
@@ -40,26 +51,36 @@ def _gloo_reduce_op(op: Op, dtype):
 
 
 def allreduce(x, op, comm):
+    _check_bitwise_dtype(op, x, "allreduce")
     with debug_timer("Allreduce", comm.rank, f"{x.numel()} items"):
         out = x.clone().contiguous()
         if comm.size == 1:
             if op is Op.AVG:
                 return out
             return out
+        # gloo refuses bitwise ops on bool tensors; ride a uint8 view
+        # (bitwise on 0/1 == logical)
+        as_bool = x.dtype == torch.bool and op in _BITWISE_OPS
+        if as_bool:
+            out = out.to(torch.uint8)
         gop = _gloo_reduce_op(op, x.dtype)
         if gop is None:  # AVG
             dist.all_reduce(out, op=dist.ReduceOp.SUM, group=comm.gloo_group)
             out = out / comm.size
         else:
             dist.all_reduce(out, op=gop, group=comm.gloo_group)
-        return out
+        return out.to(torch.bool) if as_bool else out
 
 
 def reduce(x, op, root, comm):
+    _check_bitwise_dtype(op, x, "reduce")
     with debug_timer("Reduce", comm.rank, f"{x.numel()} items"):
         out = x.clone().contiguous()
         if comm.size == 1:
             return out if comm.rank == root else None
+        as_bool = x.dtype == torch.bool and op in _BITWISE_OPS
+        if as_bool:
+            out = out.to(torch.uint8)
         gop = _gloo_reduce_op(op, x.dtype)
         if gop is None:
             dist.reduce(
@@ -72,7 +93,9 @@ def reduce(x, op, root, comm):
             dist.reduce(
                 out, dst=comm.global_rank(root), op=gop, group=comm.gloo_group
             )
-        return out if comm.rank == root else None
+        if comm.rank != root:
+            return None
+        return out.to(torch.bool) if as_bool else out
 
 
 def allgather(x, comm):
@@ -163,6 +186,7 @@ def scan(x, op, comm):
     Implemented as a rank chain (recv partial from r-1, combine, forward to
     r+1) — the same dataflow the GPU ring uses with the HIP combine kernel.
     """
+    _check_bitwise_dtype(op, x, "scan")
     with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
         out = x.clone().contiguous()
         if comm.size == 1:

@@ -21,7 +21,7 @@ Dtype strategy:
 
 import torch
 
-from ..ops.reduce_ops import Op, RCCL_OP_ENUM
+from ..ops.reduce_ops import Op, RCCL_OP_ENUM, BITWISE_OP_ENUM
 from ..utils.status import ANY_SOURCE, ANY_TAG
 from ..utils.logging import debug_timer, get_logging
 
@@ -71,10 +71,20 @@ def _bytes(x):
 
 
 def _check_op(op, x, op_name):
+    if op in BITWISE_OP_ENUM:
+        # bitwise rides the HIP-combine p2p compositions (scan ring /
+        # scan+bcast), valid for integer and bool dtypes only
+        if x.is_floating_point() or x.is_complex():
+            raise ValueError(
+                f"{op_name}: {op} requires an integer or bool dtype, "
+                f"got {x.dtype}"
+            )
+        return
     if op not in RCCL_OP_ENUM:
         raise ValueError(
             f"{op_name}: reduction {op} is not supported on the RCCL "
-            f"backend (supported: SUM, PROD, MIN, MAX, AVG)"
+            f"backend (supported: SUM, PROD, MIN, MAX, AVG, and "
+            f"BAND/BOR/BXOR on integer dtypes)"
         )
     if x.is_complex() and op not in (Op.SUM, Op.AVG):
         raise ValueError(f"{op_name}: {op} is undefined for complex dtypes")
@@ -89,6 +99,8 @@ def _reduction_view(x, op, op_name):
     if x.is_complex():
         return torch.view_as_real(x), op, None
     if x.dtype == torch.bool:
+        if op in BITWISE_OP_ENUM:  # bitwise on bool == logical, via uint8
+            return x.to(torch.uint8), op, lambda t: t.to(torch.bool)
         if op not in _BOOL_OP:
             raise ValueError(f"{op_name}: {op} undefined for bool")
         return x.to(torch.uint8), _BOOL_OP[op], lambda t: t.to(torch.bool)
@@ -101,6 +113,13 @@ def allreduce(x, op, comm):
     _check_op(op, x, "allreduce")
     if x.numel() == 0:
         return x.clone()
+    if op in BITWISE_OP_ENUM:
+        # RCCL has no bitwise reductions: inclusive scan (ring + HIP
+        # combine kernel) leaves the full reduction on the last rank,
+        # then broadcast it.  O(P) latency; exact (ops are associative
+        # and commutative, order is deterministic).
+        full = scan(x, op, comm)
+        return bcast(full, comm.size - 1, comm)
     with debug_timer("Allreduce", comm.rank, f"{x.numel()} items"):
         xr, rop, post = _reduction_view(x.contiguous(), op, "allreduce")
         out = torch.empty_like(xr)
@@ -114,6 +133,10 @@ def allreduce(x, op, comm):
 
 def reduce(x, op, root, comm):
     _check_op(op, x, "reduce")
+    if op in BITWISE_OP_ENUM:
+        out = scan(x, op, comm)
+        out = bcast(out, comm.size - 1, comm)  # see allreduce
+        return out if comm.rank == root else None
     with debug_timer("Reduce", comm.rank, f"{x.numel()} items"):
         xr, rop, post = _reduction_view(x.contiguous(), op, "reduce")
         out = torch.empty_like(xr)
@@ -135,8 +158,12 @@ def scan(x, op, comm):
         raise ValueError("scan: AVG is not a valid scan operator")
     with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
         xr, rop, post = _reduction_view(x.contiguous(), op, "scan")
-        out = torch.empty_like(xr)
-        ext().scan(out, xr, RCCL_OP_ENUM[rop], _handle(comm))
+        # the op code only ever reaches the HIP combine kernel in the
+        # scan ring (bridge.cpp scan), so kernel-only bitwise codes are
+        # legal here
+        code = (BITWISE_OP_ENUM[rop] if rop in BITWISE_OP_ENUM
+                else RCCL_OP_ENUM[rop])
+        ext().scan(out, xr, code, _handle(comm))
         if post is not None:
             out = post(out)
         if x.is_complex():

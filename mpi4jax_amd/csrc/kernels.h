@@ -20,6 +20,11 @@ enum OpCode : int {  // matches ncclRedOp_t for the first four
   OPC_PROD = 1,
   OPC_MAX = 2,
   OPC_MIN = 3,
+  // kernel-only codes (never passed to an RCCL collective — RCCL has no
+  // bitwise reductions; these ride the p2p+combine compositions)
+  OPC_BAND = 5,
+  OPC_BOR = 6,
+  OPC_BXOR = 7,
 };
 
 // dst[i] = a[i] (op) b[i], grid-strided, enqueued on `stream`

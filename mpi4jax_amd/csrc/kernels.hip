@@ -11,6 +11,8 @@
 #include <hip/hip_fp16.h>
 #include <hip/hip_bf16.h>
 
+#include <type_traits>
+
 #include "kernels.h"
 
 namespace {
@@ -63,7 +65,18 @@ __device__ inline A apply_op(int op, A x, A y) {
     case OPC_SUM: return x + y;
     case OPC_PROD: return x * y;
     case OPC_MAX: return x > y ? x : y;
-    default: return x < y ? x : y;  // OPC_MIN
+    case OPC_MIN: return x < y ? x : y;
+    default:
+      if constexpr (std::is_integral<A>::value) {
+        switch (op) {
+          case OPC_BAND: return x & y;
+          case OPC_BOR: return x | y;
+          default: return x ^ y;  // OPC_BXOR
+        }
+      } else {
+        // bitwise on floating dtypes is rejected in the Python layer
+        return x;
+      }
   }
 }
 

@@ -7,8 +7,10 @@ standalone: it defines its own :class:`Op` enum; if mpi4py happens to be
 installed, its ops are transparently mapped for drop-in compatibility.
 
 SUM/PROD/MIN/MAX (and AVG, an RCCL extra) run natively on both backends.
-The bitwise ops BAND/BOR/BXOR are CPU-only (gloo supports them; RCCL does
-not — a HIP combine kernel could add them, tracked as a gap).
+The bitwise ops BAND/BOR/BXOR (integer/bool dtypes) run on gloo natively
+and on the GPU through the HIP combine kernel over p2p chains (RCCL has
+no bitwise reductions): scan is the ring it already uses, and
+allreduce/reduce compose scan with a broadcast of the final prefix.
 """
 
 import enum
@@ -47,6 +49,16 @@ RCCL_OP_ENUM = {
     Op.MAX: 2,
     Op.MIN: 3,
     Op.AVG: 4,
+}
+
+# kernel-only codes for the HIP combine kernel (csrc/kernels.h OpCode):
+# RCCL has no bitwise reductions, so on the GPU these ops ride the
+# p2p-chain compositions (scan ring; allreduce/reduce = scan + bcast)
+# and never reach an RCCL collective.
+BITWISE_OP_ENUM = {
+    Op.BAND: 5,
+    Op.BOR: 6,
+    Op.BXOR: 7,
 }
 
 GLOO_OP_MAP = {

@@ -544,3 +544,41 @@ def test_fused_matches_eager_closed_x(nx, ny):
         assert torch.allclose(a, b, atol=1e-4, rtol=1e-4), (
             name, (a - b).abs().max().item()
         )
+
+
+@pytest.mark.parametrize("dtype", [torch.int32, torch.uint8, torch.int64])
+def test_combine_kernel_bitwise(dtype):
+    """HIP combine kernel codes 5/6/7 (BAND/BOR/BXOR) vs plain torch."""
+    import mpi4jax_amd._rccl_C as ext
+
+    g = torch.Generator().manual_seed(3)
+    info = torch.iinfo(dtype)
+    a = torch.randint(0, min(info.max, 255), (4097,), generator=g,
+                      dtype=dtype).cuda()
+    b = torch.randint(0, min(info.max, 255), (4097,), generator=g,
+                      dtype=dtype).cuda()
+    for code, fn in ((5, torch.bitwise_and), (6, torch.bitwise_or),
+                     (7, torch.bitwise_xor)):
+        dst = torch.empty_like(a)
+        ext.combine(dst, a, b, code)
+        torch.cuda.synchronize()
+        assert torch.equal(dst, fn(a, b)), code
+
+
+def test_bitwise_reductions_gpu_world1():
+    """Bitwise allreduce/reduce/scan on the RCCL backend (world 1: all
+    three must be the identity, exercising the scan+bcast composition)."""
+    x = torch.tensor([0b1100, 0b1010, 0xFFFF], dtype=torch.int32,
+                     device="cuda")
+    for op in (m.BAND, m.BOR, m.BXOR):
+        assert torch.equal(m.allreduce(x, op), x)
+        assert torch.equal(m.scan(x, op), x)
+        assert torch.equal(m.reduce(x, op, root=0), x)
+    b = torch.tensor([True, False], device="cuda")
+    assert torch.equal(m.allreduce(b, m.BAND), b)
+    assert torch.equal(m.allreduce(b, m.BOR), b)
+    # int16 rides the int32 upcast view
+    s = torch.tensor([-2, 7], dtype=torch.int16, device="cuda")
+    assert torch.equal(m.allreduce(s, m.BXOR), s)
+    with pytest.raises(ValueError, match="integer"):
+        m.allreduce(torch.zeros(2, device="cuda"), m.BAND)

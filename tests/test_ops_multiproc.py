@@ -258,3 +258,53 @@ def _usage_global_sum(rank, ws):
 
 def test_usage_example_world4():
     run_multiproc(_usage_global_sum, 4)
+
+
+def _bitwise(rank, ws):
+    # per-rank distinct bit patterns
+    x = torch.tensor([0b1100, 0b1010, 0b0110], dtype=torch.int32) << rank
+    parts = [torch.tensor([0b1100, 0b1010, 0b0110], dtype=torch.int32) << r
+             for r in range(ws)]
+
+    def fold(op, seq):
+        out = seq[0].clone()
+        for t in seq[1:]:
+            out = op(out, t)
+        return out
+
+    assert torch.equal(m.allreduce(x, m.BAND), fold(torch.bitwise_and,
+                                                    parts))
+    assert torch.equal(m.allreduce(x, m.BOR), fold(torch.bitwise_or,
+                                                   parts))
+    assert torch.equal(m.allreduce(x, m.BXOR), fold(torch.bitwise_xor,
+                                                    parts))
+
+    # inclusive prefix
+    y = m.scan(x, m.BOR)
+    assert torch.equal(y, fold(torch.bitwise_or, parts[:rank + 1])), rank
+
+    # reduce to root 0
+    y = m.reduce(x, m.BXOR, root=0)
+    if rank == 0:
+        assert torch.equal(y, fold(torch.bitwise_xor, parts))
+    else:
+        assert torch.equal(y, x)  # non-root passthrough
+
+    # bool: bitwise == logical
+    b = torch.tensor([rank == 0, True, False])
+    assert torch.equal(m.allreduce(b, m.BAND),
+                       torch.tensor([ws == 1, True, False]))
+    assert torch.equal(m.allreduce(b, m.BOR),
+                       torch.tensor([True, True, False]))
+
+    # float dtype must be rejected
+    with pytest.raises(ValueError):
+        m.allreduce(torch.zeros(2), m.BAND)
+
+
+def test_bitwise_reductions():
+    run_multiproc(_bitwise, 2)
+
+
+def test_bitwise_reductions_world4():
+    run_multiproc(_bitwise, 4)
