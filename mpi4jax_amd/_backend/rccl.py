@@ -158,6 +158,7 @@ def scan(x, op, comm):
         raise ValueError("scan: AVG is not a valid scan operator")
     with debug_timer("Scan", comm.rank, f"{x.numel()} items"):
         xr, rop, post = _reduction_view(x.contiguous(), op, "scan")
+        out = torch.empty_like(xr)
         # the op code only ever reaches the HIP combine kernel in the
         # scan ring (bridge.cpp scan), so kernel-only bitwise codes are
         # legal here
