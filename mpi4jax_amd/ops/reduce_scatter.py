@@ -36,9 +36,13 @@ def reduce_scatter(x, op, *, comm=None, token=NOTSET):
     x = x.detach()
     if x.is_cuda:
         from .._backend import rccl
-        from ..ops.reduce_ops import RCCL_OP_ENUM
+        from ..ops.reduce_ops import RCCL_OP_ENUM, BITWISE_OP_ENUM
 
         rccl._check_op(op, x, "reduce_scatter")
+        if op in BITWISE_OP_ENUM:
+            # no RCCL bitwise: full allreduce composition, then slice
+            full = rccl.allreduce(x, op, comm)
+            return full[comm.rank].clone()
         xc = x.contiguous()
         out = torch.empty(tuple(x.shape[1:]), dtype=x.dtype, device=x.device)
         rccl.ext().reduce_scatter(out.reshape(-1), xc.reshape(-1),

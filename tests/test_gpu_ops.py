@@ -582,3 +582,10 @@ def test_bitwise_reductions_gpu_world1():
     assert torch.equal(m.allreduce(s, m.BXOR), s)
     with pytest.raises(ValueError, match="integer"):
         m.allreduce(torch.zeros(2, device="cuda"), m.BAND)
+
+
+def test_reduce_scatter_bitwise_gpu():
+    xx = torch.tensor([[0b101, 0b011]], dtype=torch.int32, device="cuda")
+    y = m.reduce_scatter(xx, m.BOR)
+    torch.cuda.synchronize()
+    assert torch.equal(y, xx[0])
