@@ -143,3 +143,16 @@ def test_eager_ops_inside_compile_graph_break_ok():
 
     x = torch.arange(4.0)
     assert torch.equal(f(x), 2 * x + 1)
+
+
+def test_compiled_bitwise_allreduce():
+    """Bitwise ops trace through the custom-op layer (static op arg)."""
+    import mpi4jax_amd as m
+    from mpi4jax_amd import jit_ops
+
+    @torch.compile(fullgraph=True)
+    def f(x):
+        return jit_ops.allreduce(x, m.BXOR)
+
+    x = torch.tensor([0b101, 0b011], dtype=torch.int32)
+    assert torch.equal(f(x), x)  # world 1: identity
