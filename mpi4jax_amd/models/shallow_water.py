@@ -14,9 +14,10 @@ day is the headline metric.
 
 Two step implementations share semantics: an eager torch-op path (CPU
 and reference/debugging) and the fused MI355X path — two or three CDNA4
-kernels (stage 18: tendencies+update; stage 17: friction; optional scalar
-fallbacks) plus a single-group halo exchange, with hipGraph capture for
-multistep replay.
+kernels (stage 19: tendencies+update; stage 27: friction; selectable
+variants and scalar fallbacks) plus a one-native-call halo exchange
+(merged pack launch, single RCCL group, merged unpack launch), with
+validated hipGraph capture for multistep replay at any world size.
 """
 
 import math
@@ -227,11 +228,12 @@ class ShallowWater:
         stages — static per (dtype, env) so it is resolved once at buffer
         init, not per step.
 
-        float32 runs the vectorized (float4) stage kernels; f64 and the
-        MPI4JAX_AMD_SW_NOVEC escape hatch run the scalar ones.  Stage 8
-        (fully merged, derived fields in-register) measured slower than
-        the two-pass kernels — kept for reference, selectable via
-        MPI4JAX_AMD_SW_MERGED.
+        float32 runs the vectorized stage kernels (2 columns/thread by
+        default — the measured-fastest occupancy point); f64 and the
+        MPI4JAX_AMD_SW_NOVEC escape hatch run the scalar ones.  The
+        other variants (4-col, nontemporal, two-pass, fully-merged
+        scalar) stay selectable for measurement — see
+        profiles/README.md for the numbers behind the default.
         """
         import os
 
