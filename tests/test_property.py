@@ -49,3 +49,37 @@ def test_allreduce_grad_identity(shape, dtype, seed):
     g = make(shape, dtype, seed + 1)
     m.allreduce(x, m.SUM).backward(g)
     assert torch.equal(x.grad, g)
+
+
+INT_DTYPES = [torch.int8, torch.int32, torch.int64, torch.uint8,
+              torch.int16, torch.bool]
+
+
+@settings(max_examples=40, deadline=None)
+@given(shape=shapes, dtype=st.sampled_from(INT_DTYPES),
+       op=st.sampled_from(["band", "bor", "bxor"]),
+       seed=st.integers(0, 2**31 - 1))
+def test_bitwise_size1_identity(shape, dtype, op, seed):
+    """Bitwise reductions at size 1 are the identity for every integer
+    dtype, and never mutate the input."""
+    g = torch.Generator().manual_seed(seed)
+    if dtype == torch.bool:
+        x = torch.randint(0, 2, shape, generator=g).to(torch.bool)
+    else:
+        lo = 0 if dtype == torch.uint8 else -9
+        x = torch.randint(lo, 9, shape, generator=g).to(dtype)
+    ref = x.clone()
+    o = m.Op(op)
+    assert torch.equal(m.allreduce(x, o), x)
+    assert torch.equal(m.scan(x, o), x)
+    assert torch.equal(x, ref)
+
+
+@settings(max_examples=20, deadline=None)
+@given(dtype=st.sampled_from([torch.float32, torch.float16]),
+       op=st.sampled_from(["band", "bor", "bxor"]))
+def test_bitwise_rejects_float(dtype, op):
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        m.allreduce(torch.zeros(3, dtype=dtype), m.Op(op))
