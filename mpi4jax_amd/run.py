@@ -31,6 +31,18 @@ def main(argv=None):
         s.bind(("127.0.0.1", 0))
         port = s.getsockname()[1]
 
+    # child processes get sys.path[0] = the USER script's directory, so a
+    # source-tree (non-installed) mpi4jax_amd would not be importable for
+    # scripts outside this tree — propagate our package location
+    import os
+
+    pkg_parent = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    existing = os.environ.get("PYTHONPATH", "")
+    if pkg_parent not in existing.split(os.pathsep):
+        os.environ["PYTHONPATH"] = (
+            pkg_parent + (os.pathsep + existing if existing else "")
+        )
+
     from torch.distributed.run import main as torchrun
 
     torchrun([

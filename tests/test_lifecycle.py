@@ -121,3 +121,26 @@ def test_launcher_runs_script():
     )
     assert res.returncode == 0, res.stderr + res.stdout
     assert "SUM=3.0" in res.stdout
+
+
+def test_run_launcher_out_of_tree_script(tmp_path):
+    """python -m mpi4jax_amd.run must make the source-tree package
+    importable for user scripts living anywhere (the child's sys.path[0]
+    is the script's own directory)."""
+    import subprocess
+    import sys
+
+    script = tmp_path / "user_script.py"
+    script.write_text(
+        "import torch, mpi4jax_amd as m\n"
+        "m.init()\n"
+        "y = m.allreduce(torch.ones(2), m.SUM)\n"
+        "print('RESULT', m.get_world().rank, y.sum().item(), flush=True)\n"
+    )
+    res = subprocess.run(
+        [sys.executable, "-m", "mpi4jax_amd.run", "-n", "2", str(script)],
+        capture_output=True, text=True, timeout=240,
+    )
+    assert res.returncode == 0, res.stdout + res.stderr
+    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("RESULT")]
+    assert sorted(lines) == ["RESULT 0 4.0", "RESULT 1 4.0"], res.stdout
