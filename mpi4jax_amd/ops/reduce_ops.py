@@ -73,9 +73,15 @@ GLOO_OP_MAP = {
 
 
 def resolve_op(op, op_name="reduction"):
-    """Accept an Op, or an mpi4py Op if mpi4py is importable."""
+    """Accept an Op, its string name (as jit_ops does), or an mpi4py Op
+    if mpi4py is importable."""
     if isinstance(op, Op):
         return op
+    if isinstance(op, str):
+        try:
+            return Op(op.lower())
+        except ValueError:
+            pass  # fall through to the uniform TypeError below
     # optional mpi4py compatibility
     try:
         from mpi4py import MPI  # noqa

@@ -56,3 +56,12 @@ def test_sendrecv_device_mismatch():
     with pytest.raises(ValueError, match="same device"):
         m.sendrecv(torch.zeros(3), torch.zeros(3, device="cuda"),
                    source=0, dest=0)
+
+
+def test_string_op_names():
+    """Eager ops accept string op names, same spelling as jit_ops."""
+    x = torch.ones(3)
+    assert torch.equal(m.allreduce(x, "sum"), x)
+    assert torch.equal(m.scan(x, "MAX"), x)
+    with pytest.raises(TypeError):
+        m.allreduce(x, "not-an-op")
