@@ -68,7 +68,7 @@ def bench_allreduce(comm, device, steps=20, warmup=5):
 
 def bench_bisection(comm, device, steps=10, warmup=3, gib_per_rank=1):
     n = comm.size
-    nbytes = gib_per_rank * 1024 * 1024 * 1024
+    nbytes = int(gib_per_rank * 1024 * 1024 * 1024)
     elems = nbytes // 2
     x = torch.randn(elems, device=device).to(torch.bfloat16)
     xa = x.reshape(n, -1)
@@ -138,7 +138,7 @@ def main():
     p.add_argument("which", nargs="?", default="all",
                    choices=["all", "allreduce", "bisection", "grad",
                             "reduce_scatter"])
-    p.add_argument("--gib", type=int, default=1)
+    p.add_argument("--gib", type=float, default=1.0)
     args = p.parse_args()
     m.init()
     comm = m.get_world()
