@@ -120,7 +120,9 @@ def test_bad_rank(x):
 
 def test_bad_op(x):
     with pytest.raises(TypeError):
-        m.allreduce(x, "sum")
+        m.allreduce(x, "not-a-reduction")
+    with pytest.raises(TypeError):
+        m.allreduce(x, 123)
 
 
 def test_capability_probes():
