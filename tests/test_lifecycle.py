@@ -131,16 +131,19 @@ def test_run_launcher_out_of_tree_script(tmp_path):
     import sys
 
     script = tmp_path / "user_script.py"
+    # per-rank result files: rank stdout gets interleaved by the launcher
     script.write_text(
-        "import torch, mpi4jax_amd as m\n"
+        "import pathlib, torch, mpi4jax_amd as m\n"
         "m.init()\n"
         "y = m.allreduce(torch.ones(2), m.SUM)\n"
-        "print('RESULT', m.get_world().rank, y.sum().item(), flush=True)\n"
+        "r = m.get_world().rank\n"
+        f"pathlib.Path(r'{tmp_path}', f'out_{{r}}.txt')"
+        ".write_text(str(y.sum().item()))\n"
     )
     res = subprocess.run(
         [sys.executable, "-m", "mpi4jax_amd.run", "-n", "2", str(script)],
         capture_output=True, text=True, timeout=240,
     )
     assert res.returncode == 0, res.stdout + res.stderr
-    lines = [ln for ln in res.stdout.splitlines() if ln.startswith("RESULT")]
-    assert sorted(lines) == ["RESULT 0 4.0", "RESULT 1 4.0"], res.stdout
+    for r in (0, 1):
+        assert (tmp_path / f"out_{r}.txt").read_text() == "4.0", res.stdout
