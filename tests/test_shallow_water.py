@@ -266,3 +266,27 @@ def test_exchange_cache_flat_encoding():
             # no remote peers at world 1 => comm never resolved
             if size == 1:
                 assert cache["comm_id"] == -1
+
+
+def _agree_worker(rank, ws):
+    import torch
+
+    from mpi4jax_amd.models import ShallowWater
+
+    sw = ShallowWater(nx=12, ny=6, device="cpu")
+    # unanimous yes
+    assert sw._all_ranks_agree(True) is True
+    # one dissenter -> everyone sees False (all-or-none)
+    assert sw._all_ranks_agree(rank != 1) is False
+    # unanimous no
+    assert sw._all_ranks_agree(False) is False
+    assert isinstance(sw._all_ranks_agree(torch.tensor(True)), bool)
+
+
+def test_graph_adoption_agreement_cross_rank():
+    """The hipGraph validate-then-adopt protocol requires all ranks to
+    reach the same decision; its agreement reduction must be all-or-none
+    across ranks (gloo analog of the RCCL path)."""
+    from tests._mp import run_multiproc
+
+    run_multiproc(_agree_worker, 2)
