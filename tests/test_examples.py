@@ -39,3 +39,13 @@ def test_distributed_cg_example():
     )
     assert res.returncode == 0, res.stderr
     assert "iterations" in res.stdout
+
+
+def test_tensor_parallel_example_two_ranks():
+    res = subprocess.run(
+        [sys.executable, "-m", "mpi4jax_amd.run", "-n", "2",
+         "examples/tensor_parallel_mlp.py", "--dim", "64", "--steps", "6"],
+        capture_output=True, text=True, timeout=300, env=ENV, cwd=REPO,
+    )
+    assert res.returncode == 0, res.stderr + res.stdout
+    assert "OK: 6 TP steps over 2 rank(s)" in res.stdout
