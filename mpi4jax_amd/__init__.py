@@ -55,6 +55,12 @@ from ._backend import (  # noqa: F401
 from .utils.logging import set_logging, get_logging  # noqa: F401
 
 from .ops import jit_ops  # noqa: F401
+
+# make `import mpi4jax_amd.jit_ops` work too (it lives under ops/)
+import sys as _sys
+
+_sys.modules[__name__ + ".jit_ops"] = jit_ops
+del _sys
 from .ops import (  # noqa: F401
     allgather,
     allreduce,

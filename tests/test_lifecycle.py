@@ -147,3 +147,12 @@ def test_run_launcher_out_of_tree_script(tmp_path):
     assert res.returncode == 0, res.stdout + res.stderr
     for r in (0, 1):
         assert (tmp_path / f"out_{r}.txt").read_text() == "4.0", res.stdout
+
+
+def test_jit_ops_importable_as_submodule():
+    import importlib
+
+    mod = importlib.import_module("mpi4jax_amd.jit_ops")
+    import mpi4jax_amd as m
+
+    assert mod is m.jit_ops
