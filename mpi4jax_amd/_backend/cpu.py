@@ -7,6 +7,8 @@ SURVEY.md §4).  All functions take comm-relative ranks and NEVER mutate
 their inputs (the reference's immutability contract).
 """
 
+import weakref
+
 import torch
 import torch.distributed as dist
 
@@ -16,11 +18,12 @@ from ..utils.logging import debug_timer
 
 # Per-communicator FIFO queues for same-rank send->recv (self-messaging).
 # MPI supports buffered self sends; gloo does not, so we emulate locally.
-_SELF_QUEUES = {}
+# Weak-keyed so a GC'd communicator can never alias a new one's queue.
+_SELF_QUEUES = weakref.WeakKeyDictionary()
 
 
 def _self_queue(comm):
-    return _SELF_QUEUES.setdefault(id(comm), [])
+    return _SELF_QUEUES.setdefault(comm, [])
 
 
 def _wire(t):

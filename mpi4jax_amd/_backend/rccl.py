@@ -19,6 +19,8 @@ Dtype strategy:
   int16 upcasts to int32 on-stream (RCCL has no int16).
 """
 
+import weakref
+
 import torch
 
 from ..ops.reduce_ops import Op, RCCL_OP_ENUM, BITWISE_OP_ENUM
@@ -31,7 +33,8 @@ _EXT_ERR = None
 # per-comm FIFO for unmatched self-sends: an ncclSend to self outside a
 # group would block the stream forever, so self p2p short-circuits to a
 # stream-ordered device copy (MPI buffered-self-send semantics).
-_SELF_QUEUES = {}
+# Weak-keyed so a GC'd communicator can never alias a new one's queue.
+_SELF_QUEUES = weakref.WeakKeyDictionary()
 
 
 def ext():
@@ -227,7 +230,7 @@ def scatter(x, root, comm):
 
 
 def _self_queue(comm):
-    return _SELF_QUEUES.setdefault(id(comm), [])
+    return _SELF_QUEUES.setdefault(comm, [])
 
 
 def send(x, dest, tag, comm):

@@ -36,6 +36,7 @@ from .._backend import backend_for
 # int key <-> Communicator registry (keys are stable for a process life)
 _COMMS = {}
 _KEYS = {}
+_NEXT_KEY = [1]
 
 
 def comm_key(comm) -> int:
@@ -49,7 +50,8 @@ def comm_key(comm) -> int:
     """
     comm = resolve_comm(comm)
     if id(comm) not in _KEYS:
-        key = len(_COMMS) + 1
+        key = _NEXT_KEY[0]
+        _NEXT_KEY[0] += 1
         _COMMS[key] = comm
         _KEYS[id(comm)] = key
     return _KEYS[id(comm)]
@@ -71,7 +73,15 @@ def _key_of(comm) -> int:
 
 
 def _comm(key: int) -> Communicator:
-    return _COMMS[key]
+    c = _COMMS.get(key)
+    if c is None:
+        # finalize() cleared the registry but a cached compiled graph still
+        # holds this baked key: remap it to the (re-initialized) default
+        # communicator so old graphs keep working after finalize()+init().
+        c = resolve_comm(None)
+        _COMMS[key] = c
+        _KEYS[id(c)] = key
+    return c
 
 
 _OPS = {o.value: o for o in Op}
@@ -254,7 +264,9 @@ def _gather(x: torch.Tensor, root: int, key: int) -> torch.Tensor:
     comm = _comm(key)
     out = backend_for(x).gather(x.contiguous(), root, comm)
     if out is None:  # non-root: keep a static output shape for the graph
-        out = x.new_empty((comm.size,) + tuple(x.shape))
+        # zero-filled, not uninitialized — reading it is still meaningless
+        # on non-root, but the value is at least deterministic
+        out = x.new_zeros((comm.size,) + tuple(x.shape))
     return out
 
 
@@ -264,7 +276,7 @@ def _(x, root, key):
 
 
 def gather(x, root, *, comm=None):
-    """Root gets ``(nproc, *shape)``; other ranks get an undefined tensor
+    """Root gets ``(nproc, *shape)``; other ranks get a zero-filled tensor
     of that shape (static shapes are required inside a compiled graph —
     the eager op's non-root input passthrough does not translate)."""
     return _gather(x, root, _key_of(comm))

@@ -291,10 +291,15 @@ def flush():
             rccl.ext().check_async_errors()
     except ImportError:
         pass
-    except RuntimeError:
-        raise
     except Exception:
-        pass
+        # an async comm failure at exit must never vanish silently
+        import sys
+        import traceback
+
+        print("mpi4jax_amd.flush(): pending communication failed:",
+              file=sys.stderr)
+        traceback.print_exc()
+        raise
 
 
 def finalize():
@@ -309,6 +314,20 @@ def finalize():
                 rccl.ext().destroy_all_comms()
         except Exception:
             pass
+        # destroy_all_comms freed every native handle: drop the stale ids
+        # so a later re-init() creates fresh ones instead of dangling, and
+        # clear the jit_ops registry (cached compiled graphs re-resolve
+        # the default key at next call because _DEFAULT_COMM resets too).
+        from ..ops import jit_ops
+
+        for c in list(jit_ops._COMMS.values()):
+            c._rccl_id = None
+        jit_ops._COMMS.clear()
+        jit_ops._KEYS.clear()
+        if _WORLD is not None:
+            _WORLD._rccl_id = None
+        if _DEFAULT_COMM is not None:
+            _DEFAULT_COMM._rccl_id = None
         _WORLD = None
         _DEFAULT_COMM = None
         if dist.is_initialized():
