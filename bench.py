@@ -1,14 +1,27 @@
-"""Flagship benchmark: shallow-water solver steps/sec on MI355X.
+"""Flagship benchmark: shallow-water steps/sec + the BASELINE comm configs.
 
 Measures the reference's headline benchmark (BASELINE.json / BASELINE.md):
 the shallow-water demo on a (3600, 1800) float32 domain
 (``/root/reference/docs/shallow-water.rst:49-52``), reported as whole-job
-steps/sec (strong scaling: fixed global domain decomposed over N GPUs), plus
-the in-stream 256 MiB bf16 allreduce bandwidth when N > 1.
+steps/sec (strong scaling: fixed global domain decomposed over N GPUs),
+plus every communication config BASELINE.json names, each labeled with
+n_gpus so the driver's multi-GPU run produces the complete headline:
+
+* config #2 — 256 MiB bf16 allreduce, enqueued inside a captured hipGraph
+  (the "in-jit" analog; falls back to eager enqueue with ``in_graph:
+  false`` if capture fails on the box).  At n_gpus=1 this is a local
+  device copy, NOT communication — labeled as such.
+* config #4 — alltoall + allgather at 1 GiB/rank (xGMI bisection).
+* config #5 — torch.autograd backward through allreduce(SUM) bf16.
 
 Launch (driver contract):
     python bench.py --gpus N --steps K --warmup W
 N>1 runs under torch.distributed.run with one rank per GPU over RCCL.
+
+``--preflight`` instead validates the N>1 machinery end-to-end over the
+gloo bootstrap plane (CPU): decomposed model steps, cross-rank halo
+schedule matching, the stepper's all-or-none graph-adoption agreement —
+so the first multi-GPU contact cannot die on schedule bugs.
 
 vs_baseline compares seconds-per-model-day against the reference's
 published GPU n=1 number (103.18 s/model-day on a Tesla P100,
@@ -29,49 +42,228 @@ from mpi4jax_amd.parallel.grid import default_dims
 REF_SEC_PER_MODEL_DAY_GPU1 = 103.18  # P100, docs/shallow-water.rst:81-83
 
 
-def measure_allreduce_gbps(steps=20, warmup=5):
-    """In-stream allreduce bandwidth, 256 MiB bf16 (BASELINE config #2)."""
-    comm = m.get_world()
-    nbytes = 256 * 1024 * 1024
+def _timed_region(fn, steps, comm):
+    """barrier+sync bracketed timing; returns max-over-ranks seconds/step."""
+    m.barrier(comm=comm)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(steps):
+        fn()
+    m.barrier(comm=comm)
+    if torch.cuda.is_available():
+        torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / steps
+    dt_t = torch.tensor([dt], dtype=torch.float64)
+    return m.allreduce(dt_t, m.MAX, comm=comm).item()
+
+
+def bench_allreduce_graph(comm, nbytes=256 * 1024 * 1024, iters=20,
+                          warmup=5):
+    """BASELINE config #2: 256 MiB bf16 allreduce inside a hipGraph."""
     n = nbytes // 2
     x = torch.randn(n, device="cuda").to(torch.bfloat16)
-    m.allreduce(x, m.SUM)  # creates the default comm's RCCL communicator
-    # register the persistent input so RCCL can use zero-copy protocols
-    try:
+    m.allreduce(x, m.SUM, comm=comm)  # comm init + warm path
+    try:  # register the persistent buffer for RCCL zero-copy protocols
         from mpi4jax_amd._backend import rccl
-        from mpi4jax_amd.parallel.comm import get_default_comm
 
-        rccl.ext().comm_register(get_default_comm().rccl_handle(), x)
+        rccl.ext().comm_register(comm.rccl_handle(), x)
     except Exception:
         pass
     for _ in range(warmup):
-        m.allreduce(x, m.SUM)
-    m.barrier()
+        m.allreduce(x, m.SUM, comm=comm)
     torch.cuda.synchronize()
-    t0 = time.perf_counter()
-    for _ in range(steps):
-        y = m.allreduce(x, m.SUM)
-    m.barrier()
-    torch.cuda.synchronize()
-    dt = (time.perf_counter() - t0) / steps
+
+    in_graph = True
+    per_replay = 4
+    try:
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            for _ in range(per_replay):
+                m.allreduce(x, m.SUM, comm=comm)
+        g.replay()
+        torch.cuda.synchronize()
+        step = g.replay
+    except Exception:
+        in_graph = False
+        per_replay = 1
+        torch.cuda.synchronize()
+
+        def step():
+            m.allreduce(x, m.SUM, comm=comm)
+
+    dt = _timed_region(step, iters, comm) / per_replay
     size = comm.size
     algbw = nbytes / dt / 1e9
     busbw = algbw * (2 * (size - 1) / size) if size > 1 else algbw
-    return {"algbw_GBps": round(algbw, 2), "busbw_GBps": round(busbw, 2),
-            "bytes": nbytes, "dtype": "bf16"}
+    out = {"algbw_GBps": round(algbw, 2), "busbw_GBps": round(busbw, 2),
+           "bytes": nbytes, "dtype": "bf16", "in_graph": in_graph,
+           "n_gpus": size}
+    if size == 1:
+        out["note"] = "single rank: local device copy, NOT communication"
+    return out
+
+
+def bench_alltoall_allgather(comm, per_rank_bytes=1 << 30, iters=10,
+                             warmup=3):
+    """BASELINE config #4: alltoall + allgather, 1 GiB/rank, bf16."""
+    size = comm.size
+    res = {}
+
+    n = per_rank_bytes // 2 // size * size  # divisible by nproc
+    x = torch.randn(n, device="cuda").to(torch.bfloat16).reshape(size, -1)
+    m.alltoall(x, comm=comm)
+    for _ in range(warmup):
+        m.alltoall(x, comm=comm)
+    dt = _timed_region(lambda: m.alltoall(x, comm=comm), iters, comm)
+    nbytes = n * 2
+    algbw = nbytes / dt / 1e9
+    res["alltoall"] = {
+        "algbw_GBps": round(algbw, 2),
+        "busbw_GBps": round(algbw * (size - 1) / size, 2),
+        "bytes_per_rank": nbytes, "dtype": "bf16", "n_gpus": size,
+    }
+    del x
+
+    ng = per_rank_bytes // 2 // size  # input so the gathered output is 1 GiB
+    y = torch.randn(ng, device="cuda").to(torch.bfloat16)
+    m.allgather(y, comm=comm)
+    for _ in range(warmup):
+        m.allgather(y, comm=comm)
+    dt = _timed_region(lambda: m.allgather(y, comm=comm), iters, comm)
+    out_bytes = ng * 2 * size
+    algbw = out_bytes / dt / 1e9
+    res["allgather"] = {
+        "algbw_GBps": round(algbw, 2),
+        "busbw_GBps": round(algbw * (size - 1) / size, 2),
+        "bytes_gathered": out_bytes, "dtype": "bf16", "n_gpus": size,
+    }
+    if size == 1:
+        for v in res.values():
+            v["note"] = "single rank: local device copy, NOT communication"
+    return res
+
+
+def bench_grad_allreduce(comm, nbytes=256 * 1024 * 1024, iters=10,
+                         warmup=3):
+    """BASELINE config #5: torch.autograd backward through allreduce(SUM).
+
+    The VJP of allreduce-SUM is the identity (reference
+    allreduce.py:152-159), so the measured region is forward comm +
+    autograd machinery — the reference's ``jax.grad`` config."""
+    n = nbytes // 2
+    w = torch.randn(n, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+
+    def step():
+        y = m.allreduce(w, m.SUM, comm=comm)
+        y.float().sum().backward()
+        w.grad = None
+
+    step()
+    for _ in range(warmup):
+        step()
+    dt = _timed_region(step, iters, comm)
+    size = comm.size
+    algbw = nbytes / dt / 1e9
+    return {"ms": round(dt * 1e3, 3), "fwd_algbw_GBps": round(algbw, 2),
+            "bytes": nbytes, "dtype": "bf16", "n_gpus": size,
+            "note": "VJP of allreduce(SUM) is the identity; timed region "
+                    "= forward comm + full autograd pass"}
+
+
+def preflight():
+    """Validate the N>1 machinery over gloo (CPU) — exits nonzero on any
+    failure.  Covers: decomposed model stepping, cross-rank halo schedule
+    send/recv matching, flat-schedule encoding, and the all-or-none
+    graph-adoption agreement reduction."""
+    import sys
+
+    from mpi4jax_amd.parallel.grid import (CartesianGrid,
+                                           halo_exchange_schedule)
+
+    m.init()
+    comm = m.get_world()
+    n = comm.size
+    dims = default_dims(n)
+    failures = []
+
+    # 1. decomposed model steps and stays finite
+    sw = ShallowWater(nx=360, ny=180, comm=comm, dims=dims, device="cpu",
+                      dtype=torch.float64)
+    state = sw.initial_conditions()
+    state = sw.step(state, first_step=True)
+    for _ in range(10):
+        state = sw.step(state)
+    if not torch.isfinite(state.h).all():
+        failures.append("model diverged at world %d" % n)
+
+    # 2. halo schedule cross-rank matching: every rank's sends to peer p
+    # must pair with p's recvs from this rank, in count and size
+    grid = CartesianGrid(comm=comm, dims=dims)
+    ny, nx = sw.ny_local, sw.nx_local
+    _, col_ops, row_ops, cor_ops, _ = halo_exchange_schedule(grid, nx, ny)
+    sends, recvs = [], []
+    for k, st, rf, sidx, ridx in col_ops:
+        if st is not None:
+            sends.append((st, "col", ny))
+        if rf is not None:
+            recvs.append((rf, "col", ny))
+    for st, rf, ridx, sidx in row_ops:
+        if st is not None:
+            sends.append((st, "row", nx - 2))
+        if rf is not None:
+            recvs.append((rf, "row", nx - 2))
+    for d, st, rf in cor_ops:
+        if st is not None:
+            sends.append((st, "cor", 1))
+        if rf is not None:
+            recvs.append((rf, "cor", 1))
+    all_sends = comm._allgather_py(sends)
+    all_recvs = comm._allgather_py(recvs)
+    for r in range(n):
+        for p in range(n):
+            s = [(kind, sz) for (to, kind, sz) in all_sends[r] if to == p]
+            q = [(kind, sz) for (frm, kind, sz) in all_recvs[p] if frm == r]
+            if s != q:
+                failures.append(
+                    f"schedule mismatch {r}->{p}: sends {s} != recvs {q}")
+
+    # 3. stepper path (eager on CPU) + agreement reduction semantics
+    advance, state = sw.make_stepper(state, steps_per_call=2)
+    state = advance()
+    if not torch.isfinite(state.h).all():
+        failures.append("stepper diverged")
+    if sw._all_ranks_agree(True) is not True:
+        failures.append("agree(True) != True")
+    if sw._all_ranks_agree(comm.rank != 0) is not False:
+        failures.append("agree(mixed) != False")
+
+    all_fail = [x for fl in comm._allgather_py(failures) for x in fl]
+    if comm.rank == 0:
+        if all_fail:
+            print("PREFLIGHT_FAILED:", all_fail, flush=True)
+        else:
+            print(f"PREFLIGHT_OK world_size={n} dims={dims}", flush=True)
+    sys.exit(1 if all_fail else 0)
 
 
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=100)
-    p.add_argument("--warmup", type=int, default=20)
+    p.add_argument("--steps", type=int, default=1000)
+    p.add_argument("--warmup", type=int, default=100)
     p.add_argument("--nx", type=int, default=3600)
     p.add_argument("--ny", type=int, default=1800)
-    p.add_argument("--allreduce-bench", action="store_true", default=True)
+    p.add_argument("--preflight", action="store_true")
+    p.add_argument("--skip-comm-bench", action="store_true",
+                   help="only the shallow-water metric")
     args = p.parse_args()
 
-    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    if args.preflight:
+        preflight()
+        return
+
     m.init()
     comm = m.get_world()
     rank = comm.rank
@@ -96,7 +288,7 @@ def main():
         state = sw.initial_conditions()
         state = sw.step(state, first_step=True)
 
-    # warmup (untimed) + hipGraph capture of a 2-step multistep
+    # warmup (untimed) + hipGraph capture of a multistep
     for _ in range(args.warmup):
         state = sw.step(state)
     spc = 10 if args.steps >= 50 else 2
@@ -129,9 +321,19 @@ def main():
     spd = sw.steps_per_model_day()
     sec_per_model_day = spd / steps_per_sec
 
-    ar = None
-    if use_gpu:
-        ar = measure_allreduce_gbps()
+    comm_bench = {}
+    if use_gpu and not args.skip_comm_bench:
+        for name, fn in (
+            ("allreduce_256MiB_bf16", lambda: bench_allreduce_graph(comm)),
+            ("bisection_1GiB_bf16",
+             lambda: bench_alltoall_allgather(comm)),
+            ("grad_allreduce_256MiB_bf16",
+             lambda: bench_grad_allreduce(comm)),
+        ):
+            try:
+                comm_bench[name] = fn()
+            except Exception as e:
+                comm_bench[name] = {"error": repr(e)}
 
     if rank == 0:
         result = {
@@ -154,7 +356,7 @@ def main():
                 "sec_per_model_day": round(sec_per_model_day, 2),
                 "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}",
                 "fused_kernels": sw.fused,
-                "allreduce_256MiB_bf16": ar,
+                **comm_bench,
             },
         }
         print(json.dumps(result), flush=True)

@@ -59,7 +59,8 @@ class ShallowWater:
 
     def __init__(self, nx=360, ny=180, dx=5e3, dy=5e3, *, comm=None,
                  dims=None, device="cpu", dtype=torch.float32,
-                 periodic_x=True, lateral_viscosity=None, fused=None):
+                 periodic_x=True, lateral_viscosity=None, fused=None,
+                 ref_friction_bug=False):
         self.comm = resolve_comm(comm)
         self.grid = CartesianGrid(self.comm, dims=dims,
                                   periodic=(False, periodic_x))
@@ -105,10 +106,21 @@ class ShallowWater:
         # CFL time step (shallow_water.py:135)
         self.dt = 0.125 * min(self.dx, self.dy) / math.sqrt(GRAVITY * DEPTH)
 
+        # ref_friction_bug: reproduce the reference demo's v-friction typo
+        # (`nu*(v_shift - u)`, /root/reference/examples/shallow_water.py:
+        # 386-391) for cross-framework trajectory validation — see
+        # docs/PARITY.md.  Eager path only.
+        self.ref_friction_bug = bool(ref_friction_bug)
+        if self.ref_friction_bug and fused:
+            raise ValueError(
+                "ref_friction_bug=True requires the eager path (fused=False)"
+            )
+
         # fused CDNA4 kernel path (GPU only): 2-3 stencil kernels + one
         # halo-exchange group per step instead of ~300 eager torch kernels
         if fused is None:
-            fused = self.device.type == "cuda"
+            fused = (self.device.type == "cuda"
+                     and not self.ref_friction_bug)
         self.fused = bool(fused) and self.device.type == "cuda"
         self._fb = None  # fused buffers
 
@@ -608,8 +620,14 @@ class ShallowWater:
                                 + (gv[_I, _I] - gv[_L, _I]) / dy)
             gu = torch.zeros_like(v)
             gv = torch.zeros_like(v)
-            gu[_I, _I] = nu * (v[_I, _R] - v[_I, _I]) / dx
-            gv[_I, _I] = nu * (v[_R, _I] - v[_I, _I]) / dy
+            if self.ref_friction_bug:
+                # the reference demo's formula verbatim (mixes v and u:
+                # examples/shallow_water.py:386-391) — compat mode only
+                gu[_I, _I] = nu * (v[_I, _R] - u[_I, _I]) / dx
+                gv[_I, _I] = nu * (v[_R, _I] - u[_I, _I]) / dy
+            else:
+                gu[_I, _I] = nu * (v[_I, _R] - v[_I, _I]) / dx
+                gv[_I, _I] = nu * (v[_R, _I] - v[_I, _I]) / dy
             gu = eb(gu, "u")
             gv = eb(gv, "v")
             v = v.clone()

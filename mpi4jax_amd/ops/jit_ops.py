@@ -113,6 +113,19 @@ torch.library.register_autograd("mpi4jax_amd::allreduce", _allreduce_bwd,
                                 setup_context=_allreduce_setup)
 
 
+def _allreduce_vmap(info, in_dims, x, op, key):
+    # the reference registers a batching rule for allreduce
+    # (allreduce.py:132-135): the op is elementwise across the batch, so
+    # reducing the whole batched tensor is the batched op; the batch dim
+    # stays where it was.  Other collectives reshape across ranks, where
+    # batching changes the wire layout — vmap over them raises torch's
+    # standard "no vmap rule" error.
+    return _allreduce(x.contiguous(), op, key), in_dims[0]
+
+
+torch.library.register_vmap("mpi4jax_amd::allreduce", _allreduce_vmap)
+
+
 def allreduce(x, op=Op.SUM, *, comm=None):
     op = op.value if isinstance(op, Op) else str(op)
     return _allreduce(x, op, _key_of(comm))

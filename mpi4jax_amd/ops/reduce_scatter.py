@@ -43,11 +43,20 @@ def reduce_scatter(x, op, *, comm=None, token=NOTSET):
             # no RCCL bitwise: full allreduce composition, then slice
             full = rccl.allreduce(x, op, comm)
             return full[comm.rank].clone()
-        xc = x.contiguous()
-        out = torch.empty(tuple(x.shape[1:]), dtype=x.dtype, device=x.device)
-        rccl.ext().reduce_scatter(out.reshape(-1), xc.reshape(-1),
-                                  RCCL_OP_ENUM[op], comm.rccl_handle())
-        return out
+        # _reduction_view maps bool onto uint8 MAX/MIN (logical OR/AND),
+        # upcasts int16 (no RCCL int16), views complex as real pairs —
+        # same dtype shims as allreduce/reduce/scan
+        xr, rop, post = rccl._reduction_view(x.contiguous(), op,
+                                             "reduce_scatter")
+        out = torch.empty(tuple(xr.shape[1:]), dtype=xr.dtype,
+                          device=xr.device)
+        rccl.ext().reduce_scatter(out.reshape(-1), xr.reshape(-1),
+                                  RCCL_OP_ENUM[rop], comm.rccl_handle())
+        if post is not None:
+            out = post(out)
+        if x.is_complex():
+            out = torch.view_as_complex(out)
+        return out.reshape(tuple(x.shape[1:]))
     # CPU: allreduce then take this rank's slice (gloo has no native
     # reduce_scatter for all our dtypes)
     from .._backend import cpu

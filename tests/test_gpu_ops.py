@@ -589,3 +589,29 @@ def test_reduce_scatter_bitwise_gpu():
     y = m.reduce_scatter(xx, m.BOR)
     torch.cuda.synchronize()
     assert torch.equal(y, xx[0])
+
+
+def test_reduce_scatter_dtype_shims():
+    """reduce_scatter must ride the same dtype shims as allreduce
+    (_reduction_view): bool -> logical OR/AND via uint8 MAX/MIN, int16 via
+    int32 upcast, complex as real pairs."""
+    b = torch.tensor([[True, False, True]], device="cuda")
+    y = m.reduce_scatter(b, m.SUM)
+    torch.cuda.synchronize()
+    assert y.dtype == torch.bool
+    assert torch.equal(y, b[0])
+    # canonical bool bytes (not e.g. 2 from uint8 arithmetic SUM)
+    assert set(y.view(torch.uint8).cpu().tolist()) <= {0, 1}
+    y = m.reduce_scatter(b, m.PROD)
+    torch.cuda.synchronize()
+    assert y.dtype == torch.bool and torch.equal(y, b[0])
+
+    i = torch.tensor([[100, -7, 32000]], dtype=torch.int16, device="cuda")
+    y = m.reduce_scatter(i, m.SUM)
+    torch.cuda.synchronize()
+    assert y.dtype == torch.int16 and torch.equal(y, i[0])
+
+    c = torch.randn(1, 5, dtype=torch.complex64, device="cuda")
+    y = m.reduce_scatter(c, m.SUM)
+    torch.cuda.synchronize()
+    assert y.dtype == torch.complex64 and torch.equal(y, c[0])

@@ -290,3 +290,29 @@ def test_graph_adoption_agreement_cross_rank():
     from tests._mp import run_multiproc
 
     run_multiproc(_agree_worker, 2)
+
+
+def test_ref_friction_bug_compat_flag():
+    """docs/PARITY.md: ref_friction_bug=True reproduces the reference
+    demo's v-friction formula (examples/shallow_water.py:386-391, which
+    mixes v and u); the default fixes it.  The two trajectories must
+    diverge (the term is nonzero) and both stay finite."""
+    import pytest
+    from mpi4jax_amd.models import ShallowWater
+
+    torch.manual_seed(0)
+    fixed = ShallowWater(nx=24, ny=12, device="cpu", dtype=torch.float64)
+    compat = ShallowWater(nx=24, ny=12, device="cpu", dtype=torch.float64,
+                          ref_friction_bug=True)
+    sf = fixed.initial_conditions()
+    sc = compat.initial_conditions()
+    assert torch.equal(sf.h, sc.h)
+    sf = fixed.step(sf, first_step=True)
+    sc = compat.step(sc, first_step=True)
+    for _ in range(5):
+        sf = fixed.step(sf)
+        sc = compat.step(sc)
+    assert torch.isfinite(sf.v).all() and torch.isfinite(sc.v).all()
+    assert not torch.equal(sf.v, sc.v)
+    with pytest.raises(ValueError):
+        ShallowWater(nx=24, ny=12, ref_friction_bug=True, fused=True)
