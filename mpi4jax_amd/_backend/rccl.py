@@ -45,10 +45,23 @@ def ext():
     if _EXT_ERR is not None:
         raise _EXT_ERR
     try:
-        from .. import _rccl_C  # type: ignore
+        import os
+
+        if os.environ.get("MPI4JAX_AMD_SW_EXT", "") == "nofma":
+            # FMA-contraction control build: device kernels compiled with
+            # -ffp-contract=off (tests/test_gpu_nofma.py)
+            from .. import _rccl_C_nofma as _rccl_C  # type: ignore
+        else:
+            from .. import _rccl_C  # type: ignore
 
         _EXT = _rccl_C
         _EXT.set_logging(get_logging())
+        # opt-in fail-fast watchdog: a wedged p2p (mismatched send/recv,
+        # dead peer) aborts all communicators and exits loudly after this
+        # many seconds instead of hanging the stream forever
+        wd = os.environ.get("MPI4JAX_AMD_WATCHDOG_SEC", "").strip()
+        if wd:
+            _EXT.set_watchdog(float(wd))
         return _EXT
     except ImportError as e:  # pragma: no cover - GPU box only
         _EXT_ERR = ImportError(
@@ -233,12 +246,36 @@ def _self_queue(comm):
     return _SELF_QUEUES.setdefault(comm, [])
 
 
+def _check_gpu_tag(tag, what):
+    """RCCL has no message envelope: remote GPU p2p is matched by enqueue
+    order, NOT by tag.  Two differently-tagged in-flight messages to one
+    peer would mismatch silently — so any non-default tag on a remote GPU
+    transfer fails loudly here (the reference transmits tags through MPI,
+    mpi_ops_common.h:354-367; self-messages keep real tag matching via
+    the local queue).  Export MPI4JAX_AMD_ALLOW_GPU_TAGS=1 to accept
+    order-based matching knowingly.
+    """
+    import os
+
+    if tag in (0, ANY_TAG):
+        return
+    if os.environ.get("MPI4JAX_AMD_ALLOW_GPU_TAGS", "") == "1":
+        return
+    raise ValueError(
+        f"{what}: tag {tag} cannot be matched on the RCCL path (no message"
+        " envelope — matching is by enqueue order). Use the default tag,"
+        " or set MPI4JAX_AMD_ALLOW_GPU_TAGS=1 to accept order-based"
+        " matching."
+    )
+
+
 def send(x, dest, tag, comm):
     with debug_timer("Send", comm.rank, f"to {dest}, tag {tag}"):
         xc = x.contiguous()
         if dest == comm.rank:
             _self_queue(comm).append((tag, xc.clone()))  # stream-ordered copy
             return
+        _check_gpu_tag(tag, "send")
         ext().send(_bytes(xc), dest, _handle(comm))
 
 
@@ -261,6 +298,7 @@ def recv(template, source, tag, comm, status):
                     _fill_status(status, source, t, out)
                     return out
             raise RuntimeError("recv from self with no matching buffered send")
+        _check_gpu_tag(tag, "recv")
         ext().recv(_bytes(out), source, _handle(comm))
         _fill_status(status, source, tag, out)
         return out
@@ -274,6 +312,10 @@ def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
         if source == comm.rank and dest == comm.rank:
             out.copy_(sc.reshape(out.shape))  # stream-ordered device copy
         else:
+            if dest != comm.rank:
+                _check_gpu_tag(sendtag, "sendrecv(sendtag)")
+            if source != comm.rank:
+                _check_gpu_tag(recvtag, "sendrecv(recvtag)")
             ext().sendrecv(_bytes(sc), _bytes(out), source, dest,
                            _handle(comm))
         _fill_status(status, source, recvtag, out)

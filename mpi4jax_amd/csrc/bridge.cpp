@@ -22,9 +22,13 @@
 #include <rocprofiler-sdk-roctx/roctx.h>
 
 #include <algorithm>
+#include <atomic>
+#include <chrono>
 #include <cstdio>
+#include <deque>
 #include <mutex>
 #include <string>
+#include <thread>
 #include <unordered_map>
 
 #include "kernels.h"
@@ -122,6 +126,81 @@ void log_enqueue(const char* op, const CommEntry& c, int64_t items) {
     std::fflush(stdout);
   }
 }
+
+// -------------------------------------------------------------- watchdog
+// Opt-in fail-fast for wedged communication (reference abort discipline:
+// mpi_ops_common.h:60-78).  RCCL enqueues asynchronously, so a mismatched
+// send/recv never fails — it silently wedges the stream forever.  With
+// MPI4JAX_AMD_WATCHDOG_SEC > 0 every collective enqueue records a stream
+// event; a monitor thread aborts all communicators and exits the process
+// (rank-tagged stderr, exit code 87) if an event is still pending past
+// the deadline.  Off by default: event record costs ~2 us per enqueue.
+struct WatchEntry {
+  hipEvent_t ev;
+  std::chrono::steady_clock::time_point deadline;
+  const char* what;  // static strings only
+  int rank;
+};
+
+std::atomic<double> g_watchdog_sec{0.0};
+std::atomic<bool> g_watch_run{false};
+// leaked on purpose: a detached monitor thread may outlive static dtors
+std::mutex* g_watch_mu = new std::mutex;
+std::deque<WatchEntry>* g_watch = new std::deque<WatchEntry>;
+
+void watch_loop() {
+  using clock = std::chrono::steady_clock;
+  while (g_watch_run.load(std::memory_order_relaxed)) {
+    std::this_thread::sleep_for(std::chrono::milliseconds(20));
+    std::lock_guard<std::mutex> lk(*g_watch_mu);
+    for (auto it = g_watch->begin(); it != g_watch->end();) {
+      hipError_t q = hipEventQuery(it->ev);
+      if (q == hipSuccess) {
+        (void)hipEventDestroy(it->ev);
+        it = g_watch->erase(it);
+        continue;
+      }
+      if (clock::now() > it->deadline) {
+        std::fprintf(stderr,
+                     "[mpi4jax_amd r%d] WATCHDOG: %s still pending after "
+                     "%.1fs — aborting all RCCL communicators and exiting "
+                     "(mismatched send/recv or a dead peer wedged the "
+                     "stream)\n",
+                     it->rank, it->what, g_watchdog_sec.load());
+        std::fflush(stderr);
+        {
+          std::lock_guard<std::mutex> ck(g_mutex);
+          for (auto& kv : g_comms) ncclCommAbort(kv.second.comm);
+        }
+        std::_Exit(87);
+      }
+      ++it;
+    }
+  }
+}
+
+void watchdog_arm(const char* what, int rank, hipStream_t stream) {
+  double t = g_watchdog_sec.load(std::memory_order_relaxed);
+  if (t <= 0) return;
+  hipEvent_t ev;
+  HIP_CHECK(hipEventCreateWithFlags(&ev, hipEventDisableTiming));
+  HIP_CHECK(hipEventRecord(ev, stream));
+  auto deadline = std::chrono::steady_clock::now() +
+                  std::chrono::duration_cast<
+                      std::chrono::steady_clock::duration>(
+                      std::chrono::duration<double>(t));
+  std::lock_guard<std::mutex> lk(*g_watch_mu);
+  g_watch->push_back({ev, deadline, what, rank});
+}
+
+void set_watchdog(double seconds) {
+  g_watchdog_sec.store(seconds);
+  if (seconds > 0 && !g_watch_run.exchange(true)) {
+    std::thread(watch_loop).detach();
+  }
+}
+
+double get_watchdog() { return g_watchdog_sec.load(); }
 
 // roctx range around every collective enqueue — shows up in
 // `rocprofv3 --marker-trace` timelines (tracing spans)
@@ -250,6 +329,7 @@ void allreduce(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
   RCCL_CHECK(ncclAllReduce(in.data_ptr(), out.data_ptr(), in.numel(),
                            nccl_dtype(in), (ncclRedOp_t)op, c.comm,
                            cur_stream()));
+  watchdog_arm("allreduce", c.rank, cur_stream());
 }
 
 void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
@@ -261,6 +341,7 @@ void reduce(at::Tensor out, at::Tensor in, int64_t op, int64_t root,
   RCCL_CHECK(ncclReduce(in.data_ptr(), out.data_ptr(), in.numel(),
                         nccl_dtype(in), (ncclRedOp_t)op, (int)root, c.comm,
                         cur_stream()));
+  watchdog_arm("reduce", c.rank, cur_stream());
 }
 
 void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
@@ -271,6 +352,7 @@ void allgather(at::Tensor out, at::Tensor in, int64_t comm_id) {
   log_enqueue("Allgather", c, in.numel());
   RCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), in.numel(),
                            nccl_dtype(in), c.comm, cur_stream()));
+  watchdog_arm("allgather", c.rank, cur_stream());
 }
 
 void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
@@ -280,6 +362,7 @@ void broadcast(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   log_enqueue("Bcast", c, in.numel());
   RCCL_CHECK(ncclBroadcast(in.data_ptr(), out.data_ptr(), in.numel(),
                            nccl_dtype(in), (int)root, c.comm, cur_stream()));
+  watchdog_arm("bcast", c.rank, cur_stream());
 }
 
 void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
@@ -293,6 +376,7 @@ void reduce_scatter(at::Tensor out, at::Tensor in, int64_t op,
   RCCL_CHECK(ncclReduceScatter(in.data_ptr(), out.data_ptr(), out.numel(),
                                nccl_dtype(in), (ncclRedOp_t)op, c.comm,
                                cur_stream()));
+  watchdog_arm("reduce_scatter", c.rank, cur_stream());
 }
 
 
@@ -349,6 +433,7 @@ void alltoall(at::Tensor out, at::Tensor in, int64_t comm_id) {
              cur_stream());
   }
   RCCL_CHECK(ncclGroupEnd());
+  watchdog_arm("alltoall", c.rank, cur_stream());
 }
 
 void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
@@ -371,6 +456,7 @@ void gather(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
     }
   }
   RCCL_CHECK(ncclGroupEnd());
+  watchdog_arm("gather", c.rank, cur_stream());
 }
 
 void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
@@ -393,6 +479,7 @@ void scatter(at::Tensor out, at::Tensor in, int64_t root, int64_t comm_id) {
   p2p_recv(out.data_ptr(), chunk, dt, esz, (int)root, c.comm,
            cur_stream());
   RCCL_CHECK(ncclGroupEnd());
+  watchdog_arm("scatter", c.rank, cur_stream());
 }
 
 void send(at::Tensor in, int64_t dest, int64_t comm_id) {
@@ -402,6 +489,7 @@ void send(at::Tensor in, int64_t dest, int64_t comm_id) {
   log_enqueue("Send", c, in.numel());
   p2p_send(in.data_ptr(), in.numel(), nccl_dtype(in), in.element_size(),
            (int)dest, c.comm, cur_stream());
+  watchdog_arm("send", c.rank, cur_stream());
 }
 
 void recv(at::Tensor out, int64_t source, int64_t comm_id) {
@@ -411,6 +499,7 @@ void recv(at::Tensor out, int64_t source, int64_t comm_id) {
   log_enqueue("Recv", c, out.numel());
   p2p_recv(out.data_ptr(), out.numel(), nccl_dtype(out),
            out.element_size(), (int)source, c.comm, cur_stream());
+  watchdog_arm("recv", c.rank, cur_stream());
 }
 
 void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
@@ -427,6 +516,7 @@ void sendrecv(at::Tensor sendbuf, at::Tensor recvbuf, int64_t source,
   p2p_recv(recvbuf.data_ptr(), recvbuf.numel(), nccl_dtype(recvbuf),
            recvbuf.element_size(), (int)source, c.comm, cur_stream());
   RCCL_CHECK(ncclGroupEnd());
+  watchdog_arm("sendrecv", c.rank, cur_stream());
 }
 
 void barrier(int64_t comm_id) {
@@ -437,6 +527,7 @@ void barrier(int64_t comm_id) {
   // 
   RCCL_CHECK(ncclAllReduce(c.barrier_buf, c.barrier_buf, 1, ncclInt32,
                            ncclSum, c.comm, cur_stream()));
+  watchdog_arm("barrier", c.rank, cur_stream());
 }
 
 // scan: ring chain with on-device combine.
@@ -465,10 +556,14 @@ void scan(at::Tensor out, at::Tensor in, int64_t op, int64_t comm_id) {
              c.rank + 1, c.comm, stream);
   }
   log_enqueue("Scan", c, in.numel());
+  watchdog_arm("scan", c.rank, stream);
 }
 
 void group_start() { RCCL_CHECK(ncclGroupStart()); }
-void group_end() { RCCL_CHECK(ncclGroupEnd()); }
+void group_end() {
+  RCCL_CHECK(ncclGroupEnd());
+  watchdog_arm("group_end", -1, cur_stream());
+}
 
 // LDS-staged strided pack/unpack (CDNA4 kernels in kernels.hip).
 // pack2d: gather a 2-D strided view into a contiguous buffer.
@@ -785,6 +880,7 @@ void sw_exchange(std::vector<at::Tensor> fields,
                        un_c[1], cor_mask ? cor_rbuf.data_ptr() : nullptr,
                        (int)cor_mask, is_double ? 1 : 0, stream);
   }
+  watchdog_arm("sw_exchange", c.rank, stream);
 }
 
 // direct access to the combine kernel (used by gpu numerics tests)
@@ -809,6 +905,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("comm_register", &comm_register);
   m.def("comm_deregister", &comm_deregister);
   m.def("set_logging", &set_logging);
+  m.def("set_watchdog", &set_watchdog);
+  m.def("get_watchdog", &get_watchdog);
   m.def("version_info", &version_info);
   m.def("allreduce", &allreduce);
   m.def("reduce", &reduce);

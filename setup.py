@@ -42,10 +42,22 @@ def _newer(target, sources):
     return any(s.stat().st_mtime > t for s in sources)
 
 
-def build_native(force=False):
+def build_native(force=False, variant=""):
+    """Build the in-tree extension.
+
+    variant="nofma" builds ``_rccl_C_nofma.so`` with the device kernels
+    compiled ``-ffp-contract=off`` — the FMA-contraction control used to
+    pin the f64 fused-vs-eager divergence to FMA alone
+    (tests/test_gpu_nofma.py); select it at runtime with
+    ``MPI4JAX_AMD_SW_EXT=nofma``.
+    """
     includes, libpaths = _torch_paths()
     py_inc = sysconfig.get_paths()["include"]
-    BUILD.mkdir(exist_ok=True)
+    suffix = f"_{variant}" if variant else ""
+    ext_name = f"_rccl_C{suffix}"
+    out_so = ROOT / "mpi4jax_amd" / f"{ext_name}.so"
+    build_dir = BUILD / variant if variant else BUILD
+    build_dir.mkdir(parents=True, exist_ok=True)
 
     common = [
         f"--offload-arch={ARCH}",
@@ -56,7 +68,7 @@ def build_native(force=False):
         "-DUSE_ROCM=1",
         "-DHIPBLAS_V2",
         "-D_GLIBCXX_USE_CXX11_ABI=1",
-        "-DTORCH_EXTENSION_NAME=_rccl_C",
+        f"-DTORCH_EXTENSION_NAME={ext_name}",
         "-DTORCH_API_INCLUDE_EXTENSION_H",
         "-fvisibility=hidden",
         "-Wno-unused-result",
@@ -67,22 +79,26 @@ def build_native(force=False):
     objs = []
     for src in ["kernels.hip", "shallow_water.hip", "bridge.cpp"]:
         sp = CSRC / src
-        op = BUILD / (src.replace(".", "_") + ".o")
+        op = build_dir / (src.replace(".", "_") + ".o")
         objs.append(op)
         deps = [sp, CSRC / "kernels.h"]
+        variant_flags = []
+        if variant == "nofma" and src.endswith(".hip"):
+            variant_flags = ["-ffp-contract=off"]
         if force or _newer(op, deps):
             extra = ["-x", "hip"] if src.endswith(".cpp") else []
             # bridge.cpp is host-only but compiled as hip for runtime hdrs
-            _run([HIPCC, "-c", *common, *inc_flags, *extra, sp, "-o", op])
+            _run([HIPCC, "-c", *common, *variant_flags, *inc_flags, *extra,
+                  sp, "-o", op])
 
-    if force or _newer(OUT_SO, objs):
+    if force or _newer(out_so, objs):
         link = [
             HIPCC,
             "-shared",
             "-fPIC",
             *objs,
             "-o",
-            OUT_SO,
+            out_so,
         ]
         for lp in libpaths:
             link += [f"-L{lp}", f"-Wl,-rpath,{lp}"]
@@ -98,12 +114,15 @@ def build_native(force=False):
             "-lamdhip64",
         ]
         _run(link)
-    print(f"built {OUT_SO}")
-    return OUT_SO
+    print(f"built {out_so}")
+    return out_so
 
 
 if __name__ == "__main__":
     if "build_ext" in sys.argv or len(sys.argv) == 1:
         build_native(force="--force" in sys.argv)
+        if "--nofma" in sys.argv or os.environ.get(
+                "MPI4JAX_AMD_BUILD_NOFMA") == "1":
+            build_native(force="--force" in sys.argv, variant="nofma")
     else:
         print(__doc__)

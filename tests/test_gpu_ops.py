@@ -615,3 +615,36 @@ def test_reduce_scatter_dtype_shims():
     y = m.reduce_scatter(c, m.SUM)
     torch.cuda.synchronize()
     assert y.dtype == torch.complex64 and torch.equal(y, c[0])
+
+
+def test_gpu_tag_hard_failure(monkeypatch):
+    """A non-default tag on a remote RCCL transfer must fail loudly — RCCL
+    has no envelope, so matching is by enqueue order and two differently-
+    tagged in-flight messages would mismatch silently.
+    Self-messages keep real tag matching via the local queue."""
+    from mpi4jax_amd._backend import rccl
+
+    class FakeComm:  # pretends we are rank 1 so peer 0 is remote
+        rank, size = 1, 2
+
+    x = torch.ones(4, device="cuda")
+    monkeypatch.delenv("MPI4JAX_AMD_ALLOW_GPU_TAGS", raising=False)
+    with pytest.raises(ValueError, match="tag 7"):
+        rccl.send(x, 0, 7, FakeComm())
+    with pytest.raises(ValueError, match="tag 9"):
+        rccl.recv(x, 0, 9, FakeComm(), None)
+    with pytest.raises(ValueError, match="tag 3"):
+        rccl.sendrecv(x, x, 0, 0, 3, -1, FakeComm(), None)
+    # the escape hatch accepts order-based matching knowingly (the check
+    # is all that runs before the native enqueue, so probe it directly)
+    monkeypatch.setenv("MPI4JAX_AMD_ALLOW_GPU_TAGS", "1")
+    rccl._check_gpu_tag(7, "send")
+    # self-messages still match tags exactly, regardless of tag value
+    monkeypatch.delenv("MPI4JAX_AMD_ALLOW_GPU_TAGS", raising=False)
+    comm = m.get_world()
+    m.send(x, 0, tag=42)
+    m.send(2 * x, 0, tag=43)
+    got43 = m.recv(x, 0, tag=43)
+    got42 = m.recv(x, 0, tag=42)
+    torch.cuda.synchronize()
+    assert got43[0].item() == 2.0 and got42[0].item() == 1.0
