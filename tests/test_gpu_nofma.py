@@ -1,66 +1,119 @@
-"""Pin the f64 fused-vs-eager divergence to FMA contraction.
+"""Pin the f64 fused-vs-eager divergence story with experiments, not
+comments.
 
 Round 1 loosened the f64 fused-vs-eager tolerance to 5e-7 after observing
-~1e-8 divergence, attributing it to hipcc contracting the stencil's
-mul+add chains into FMAs (the eager path runs each torch op as a separate
-kernel, so no cross-op contraction happens there).  This proves it: the
-``_rccl_C_nofma`` build compiles the identical kernel sources with
-``-ffp-contract=off``; under it the fused f64 trajectory must match the
-eager path BITWISE, restoring the tight-tolerance assertion for that
-variant.  Runs in a subprocess because the extension choice
-(MPI4JAX_AMD_SW_EXT) is fixed at first import.
+~1e-8 divergence and attributed it to FMA contraction.  The committed
+experiment (profiles/fma_experiment_r02.md, tools/fma_diag.py on MI355X)
+shows that attribution was WRONG in an informative way:
+
+* the ``_rccl_C_nofma`` build (identical kernels, ``-ffp-contract=off``)
+  produces trajectories BITWISE IDENTICAL to the normal build — FMA
+  contraction contributes nothing to the divergence;
+* both fused and eager paths are bitwise deterministic across runs (no
+  races);
+* the divergence appears on the very first step (u: 8e-11, v: 2e-9 at
+  128x96) and grows ~1.5e-9/step — it is evaluation-order rounding: the
+  fused kernel evaluates each tendency as one expression tree while the
+  eager path rounds after every torch op.
+
+The 5e-7 tolerance therefore covers a few hundred steps of deterministic
+order-rounding growth; these tests keep all three facts pinned.
 """
 
 import os
 import subprocess
 import sys
+import tempfile
 import textwrap
 
 import pytest
+import torch
 
 pytestmark = pytest.mark.gpu
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
-BITWISE_SCRIPT = textwrap.dedent("""
+TRAJ_SCRIPT = textwrap.dedent("""
     import os, sys, torch
-    os.environ["MPI4JAX_AMD_SW_EXT"] = "nofma"
-    sys.path.insert(0, %r)
+    sys.path.insert(0, %(repo)r)
     import mpi4jax_amd as m
     from mpi4jax_amd._backend import rccl
     m.init()
-    assert "nofma" in rccl.ext().__file__, rccl.ext().__file__
+    expect = %(expect)r
+    assert expect in rccl.ext().__file__, rccl.ext().__file__
     from mpi4jax_amd.models import ShallowWater
 
     torch.manual_seed(0)
-    kw = dict(nx=128, ny=96, device="cuda", dtype=torch.float64)
-    fused = ShallowWater(fused=True, **kw)
-    eager = ShallowWater(fused=False, **kw)
-    sf = fused.initial_conditions()
-    se = eager.initial_conditions()
-    sf = fused.step(sf, first_step=True)
-    se = eager.step(se, first_step=True)
-    for i in range(30):
-        sf = fused.step(sf)
-        se = eager.step(se)
+    sw = ShallowWater(nx=128, ny=96, device="cuda", dtype=torch.float64,
+                      fused=True)
+    s = sw.initial_conditions()
+    s = sw.step(s, first_step=True)
+    for _ in range(30):
+        s = sw.step(s)
     torch.cuda.synchronize()
-    for name in ("h", "u", "v"):
-        a, b = getattr(sf, name), getattr(se, name)
-        if not torch.equal(a, b):
-            err = (a - b).abs().max().item()
-            print("MISMATCH", name, err, flush=True)
-            sys.exit(2)
-    print("BITWISE_EQUAL", flush=True)
+    torch.save({k: getattr(s, k).cpu() for k in ("h", "u", "v")},
+               %(out)r)
+    print("TRAJ_SAVED", flush=True)
 """)
 
 
-def test_nofma_build_matches_eager_bitwise_f64():
+def _run_traj(out_path, nofma):
+    env = dict(os.environ)
+    if nofma:
+        env["MPI4JAX_AMD_SW_EXT"] = "nofma"
+    else:
+        env.pop("MPI4JAX_AMD_SW_EXT", None)
+    script = TRAJ_SCRIPT % {
+        "repo": REPO,
+        "expect": "_rccl_C_nofma.so" if nofma else "_rccl_C.so",
+        "out": out_path,
+    }
+    r = subprocess.run([sys.executable, "-c", script], cwd=REPO, env=env,
+                       capture_output=True, text=True, timeout=600)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "TRAJ_SAVED" in r.stdout
+    return torch.load(out_path)
+
+
+def test_fma_contraction_plays_no_role():
+    """The -ffp-contract=off build must reproduce the normal build
+    BITWISE over 30 fused f64 steps — proving the fused-vs-eager
+    divergence is evaluation-order rounding, not FMA."""
     if not os.path.exists(os.path.join(REPO, "mpi4jax_amd",
                                        "_rccl_C_nofma.so")):
         pytest.skip("nofma variant not built (run setup.py --nofma)")
-    r = subprocess.run(
-        [sys.executable, "-c", BITWISE_SCRIPT % REPO], cwd=REPO,
-        capture_output=True, text=True, timeout=600,
-    )
-    assert r.returncode == 0, r.stdout + r.stderr
-    assert "BITWISE_EQUAL" in r.stdout, r.stdout
+    with tempfile.TemporaryDirectory() as td:
+        a = _run_traj(os.path.join(td, "fma.pt"), nofma=False)
+        b = _run_traj(os.path.join(td, "nofma.pt"), nofma=True)
+    for k in ("h", "u", "v"):
+        assert torch.equal(a[k], b[k]), k
+
+
+def test_fused_divergence_is_deterministic_order_rounding():
+    """Both paths bitwise-deterministic across runs; fused-vs-eager
+    divergence bounded by the documented envelope at 30 steps."""
+    import mpi4jax_amd as m
+    from mpi4jax_amd.models import ShallowWater
+
+    m.init()
+
+    def traj(fused):
+        torch.manual_seed(0)
+        sw = ShallowWater(nx=128, ny=96, device="cuda",
+                          dtype=torch.float64, fused=fused)
+        s = sw.initial_conditions()
+        s = sw.step(s, first_step=True)
+        for _ in range(30):
+            s = sw.step(s)
+        torch.cuda.synchronize()
+        return {k: getattr(s, k).clone() for k in ("h", "u", "v")}
+
+    f1, f2 = traj(True), traj(True)
+    e1, e2 = traj(False), traj(False)
+    for k in ("h", "u", "v"):
+        assert torch.equal(f1[k], f2[k]), f"fused nondeterministic: {k}"
+        assert torch.equal(e1[k], e2[k]), f"eager nondeterministic: {k}"
+        err = (f1[k] - e1[k]).abs().max().item()
+        # observed ~7e-8 at 30 steps (profiles/fma_experiment_r02.md);
+        # 5e-7 is the documented envelope for this horizon
+        assert err < 5e-7, (k, err)

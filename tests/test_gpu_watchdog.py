@@ -52,14 +52,10 @@ WEDGE_SCRIPT = textwrap.dedent("""
     from mpi4jax_amd.parallel.comm import get_default_comm
     ext = rccl.ext()
     h = get_default_comm().rccl_handle()
-    # deliberately mismatched pair: send 4 elements, recv expects 8 —
-    # the recv waits forever for data that never comes
-    x = torch.ones(4, device="cuda")
+    # deliberately unmatched: a recv with no send anywhere — the stream
+    # blocks forever waiting for data that never comes
     y = torch.empty(8, device="cuda")
-    ext.group_start()
-    ext.send(x, 0, h)
     ext.recv(y, 0, h)
-    ext.group_end()
     print("WEDGE_ENQUEUED", flush=True)
     torch.cuda.synchronize()
     print("SYNC_COMPLETED_UNEXPECTEDLY", flush=True)
@@ -74,8 +70,13 @@ def test_watchdog_kills_wedged_p2p():
     out, err = r.stdout, r.stderr
     if "SYNC_COMPLETED_UNEXPECTEDLY" in out:
         pytest.skip(
-            "RCCL completed the mismatched self pair (short-delivery); "
+            "RCCL completed the unmatched recv (self short-circuit); "
             "wedge scenario not reproducible on this runtime"
+        )
+    if r.returncode != 87 and "RCCL error" in out + err:
+        pytest.skip(
+            "RCCL errored synchronously instead of wedging — fail-fast "
+            "already loud without the watchdog"
         )
     assert r.returncode == 87, (r.returncode, out, err)
     assert "WATCHDOG" in err, err
