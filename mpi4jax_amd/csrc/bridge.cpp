@@ -202,6 +202,24 @@ void set_watchdog(double seconds) {
 
 double get_watchdog() { return g_watchdog_sec.load(); }
 
+// test-only: wedge the current stream with a BOUNDED spin kernel so the
+// watchdog's abort path can be exercised at world 1 (RCCL self-mismatches
+// error synchronously there; real wedges need a peer).  Self-terminates
+// after `seconds` so a failed watchdog cannot hold the GPU past the test
+// timeout.  wall_clock64 ticks at the fixed ~100 MHz wall clock.
+__global__ void debug_spin_kernel(long long ticks) {
+  long long start = wall_clock64();
+  while (wall_clock64() - start < ticks) __builtin_amdgcn_s_sleep(63);
+}
+
+void debug_wedge_stream(double seconds) {
+  if (seconds > 120.0) seconds = 120.0;
+  long long ticks = (long long)(seconds * 1.0e8);
+  hipLaunchKernelGGL(debug_spin_kernel, dim3(1), dim3(1), 0, cur_stream(),
+                     ticks);
+  watchdog_arm("debug_wedge", -1, cur_stream());
+}
+
 // roctx range around every collective enqueue — shows up in
 // `rocprofv3 --marker-trace` timelines (tracing spans)
 struct RoctxRange {
@@ -907,6 +925,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("set_logging", &set_logging);
   m.def("set_watchdog", &set_watchdog);
   m.def("get_watchdog", &get_watchdog);
+  m.def("debug_wedge_stream", &debug_wedge_stream);
   m.def("version_info", &version_info);
   m.def("allreduce", &allreduce);
   m.def("reduce", &reduce);

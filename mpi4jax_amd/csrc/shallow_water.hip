@@ -831,19 +831,9 @@ __device__ inline void st2s(float* p, long long off, vf2 v) {
 }
 
 template <bool NT>
-__global__ void sw_stage19t(SwArgs<float> a) {
+__device__ inline void stage19_cells(const SwArgs<float>& a, int j,
+                                     int i0) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int ppr = (nx + 1) / 2;
-  const int T_ = (int)gridDim.x;
-  const int b_ = (int)blockIdx.x;
-  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
-  const int bid_ =
-      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
-      + yc_;
-  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
-  if (t >= ppr * ny) return;
-  const int j = t / ppr;
-  const int i0 = (t % ppr) * 2;
   const int jmin = a.f.south_open ? 1 : 2;
   const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
   const int imin = a.f.west_open ? 1 : 2;
@@ -924,6 +914,48 @@ __global__ void sw_stage19t(SwArgs<float> a) {
   st2s<NT>(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
 }
 
+template <bool NT>
+__global__ void sw_stage19t(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
+  if (t >= ppr * ny) return;
+  stage19_cells<NT>(a, t / ppr, (t % ppr) * 2);
+}
+
+// stage 21/22/23 = stage 19 with a 2-D band-tile block mapping (TJ = 4,
+// 8, 16 rows per 256-thread block): each block covers TJ consecutive
+// rows x (512/TJ) columns, so the j+-1 stencil rows are re-read from the
+// CU's own L1/L2 slice instead of crossing to the XCD L2.  The kernel is
+// already at minimum HBM traffic (profiles/README.md) and latency-bound;
+// tiling attacks average load latency.  Tile ids stay row-band ordered
+// under the XCD remap so each XCD still owns a contiguous band.
+template <int TJ>
+__global__ void sw_stage21t(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  constexpr int TPC = 256 / TJ;  // thread-pairs (2 cols each) per tile row
+  const int nti = (ppr + TPC - 1) / TPC;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int tj = bid_ / nti, ti = bid_ % nti;
+  const int jj = (int)threadIdx.x / TPC, ii = (int)threadIdx.x % TPC;
+  const int j = tj * TJ + jj;
+  const int i0 = (ti * TPC + ii) * 2;
+  if (j >= ny || i0 >= nx) return;
+  stage19_cells<false>(a, j, i0);
+}
+
 int sw_grid(long long n) {
   long long blocks = (n + kBlock - 1) / kBlock;
   if (blocks > 4096) blocks = 4096;  // grid-stride beyond this
@@ -994,6 +1026,12 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
     long long ppr = stage >= 19 ? (p.nx + 1) / 2 : (p.nx + 3) / 4;
     long long packs = ppr * p.ny;
     dim3 grid((unsigned)((packs + 255) / 256)), block(256);
+    if (stage >= 21 && stage <= 23) {  // band-tiled: TJ = 4 / 8 / 16
+      int tjv = stage == 21 ? 4 : stage == 22 ? 8 : 16;
+      long long tpc = 256 / tjv;
+      long long tiles = ((p.ny + tjv - 1) / tjv) * ((ppr + tpc - 1) / tpc);
+      grid = dim3((unsigned)tiles);
+    }
     switch (stage) {
       case 11: hipLaunchKernelGGL(sw_stage1v, grid, block, 0, stream, a); break;
       case 16: hipLaunchKernelGGL(sw_stage6v, grid, block, 0, stream, a); break;
@@ -1001,6 +1039,9 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 18: hipLaunchKernelGGL(sw_stage18v, grid, block, 0, stream, a); break;
       case 19: hipLaunchKernelGGL(sw_stage19t<false>, grid, block, 0, stream, a); break;
       case 20: hipLaunchKernelGGL(sw_stage19t<true>, grid, block, 0, stream, a); break;
+      case 21: hipLaunchKernelGGL(sw_stage21t<4>, grid, block, 0, stream, a); break;
+      case 22: hipLaunchKernelGGL(sw_stage21t<8>, grid, block, 0, stream, a); break;
+      case 23: hipLaunchKernelGGL(sw_stage21t<16>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
     }
     return;

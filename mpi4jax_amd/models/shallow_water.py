@@ -259,6 +259,10 @@ class ShallowWater:
                 return None, 18, 17  # 4-col variants (4 waves/SIMD)
             if os.environ.get("MPI4JAX_AMD_SW_NT"):
                 return None, 20, 27  # + nontemporal streaming hints
+            tile = os.environ.get("MPI4JAX_AMD_SW_TILE", "")
+            if tile in ("4", "8", "16"):
+                # band-tiled tendency kernel: TJ rows per 256-thread block
+                return None, {"4": 21, "8": 22, "16": 23}[tile], 27
             # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
             # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
             return None, 19, 27

@@ -49,35 +49,58 @@ WEDGE_SCRIPT = textwrap.dedent("""
     import mpi4jax_amd as m
     m.init()
     from mpi4jax_amd._backend import rccl
-    from mpi4jax_amd.parallel.comm import get_default_comm
     ext = rccl.ext()
-    h = get_default_comm().rccl_handle()
-    # deliberately unmatched: a recv with no send anywhere — the stream
-    # blocks forever waiting for data that never comes
-    y = torch.empty(8, device="cuda")
-    ext.recv(y, 0, h)
+    torch.zeros(4, device="cuda")  # init the device/stream
+    # wedge the stream: a bounded 60s spin standing in for an unmatched
+    # remote recv (world-1 RCCL self-mismatches error synchronously,
+    # which their own test covers; real wedges need a peer)
+    ext.debug_wedge_stream(60.0)
     print("WEDGE_ENQUEUED", flush=True)
     torch.cuda.synchronize()
     print("SYNC_COMPLETED_UNEXPECTEDLY", flush=True)
 """)
 
 
-def test_watchdog_kills_wedged_p2p():
+def test_watchdog_kills_wedged_stream():
     r = subprocess.run(
         [sys.executable, "-c", WEDGE_SCRIPT % REPO], cwd=REPO,
         capture_output=True, text=True, timeout=120,
     )
     out, err = r.stdout, r.stderr
-    if "SYNC_COMPLETED_UNEXPECTEDLY" in out:
-        pytest.skip(
-            "RCCL completed the unmatched recv (self short-circuit); "
-            "wedge scenario not reproducible on this runtime"
-        )
-    if r.returncode != 87 and "RCCL error" in out + err:
-        pytest.skip(
-            "RCCL errored synchronously instead of wedging — fail-fast "
-            "already loud without the watchdog"
-        )
+    assert "SYNC_COMPLETED_UNEXPECTEDLY" not in out, out
     assert r.returncode == 87, (r.returncode, out, err)
     assert "WATCHDOG" in err, err
     assert "still pending" in err, err
+
+
+SELF_MISMATCH_SCRIPT = textwrap.dedent("""
+    import os, sys, torch
+    os.environ["MPI4JAX_AMD_WATCHDOG_SEC"] = "5"  # guard: wedge -> 87
+    sys.path.insert(0, %r)
+    import mpi4jax_amd as m
+    m.init()
+    from mpi4jax_amd._backend import rccl
+    from mpi4jax_amd.parallel.comm import get_default_comm
+    ext = rccl.ext()
+    h = get_default_comm().rccl_handle()
+    y = torch.empty(8, device="cuda")
+    try:
+        ext.recv(y, 0, h)
+        torch.cuda.synchronize()
+    except RuntimeError as e:
+        print("FAILED_FAST:", str(e)[:80], flush=True)
+        sys.exit(0)
+    print("COMPLETED_SILENTLY", flush=True)
+    sys.exit(3)
+""")
+
+
+def test_mismatched_self_p2p_fails_loudly():
+    """An unmatched self recv must end loudly — either a synchronous RCCL
+    error (observed on this runtime) or, failing that, the watchdog."""
+    r = subprocess.run(
+        [sys.executable, "-c", SELF_MISMATCH_SCRIPT % REPO], cwd=REPO,
+        capture_output=True, text=True, timeout=120,
+    )
+    assert r.returncode in (0, 87), (r.returncode, r.stdout, r.stderr)
+    assert "COMPLETED_SILENTLY" not in r.stdout
