@@ -763,19 +763,9 @@ __global__ void sw_stage18v(SwArgs<float> a) {
 
 // stage 27 = stage 17 (friction Laplacian) at 2 columns/thread — same
 // occupancy rationale as stage 19 below.
-__global__ void sw_stage27v(SwArgs<float> a) {
+__device__ inline void stage27_pair(const SwArgs<float>& a, int j,
+                                    int i0) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  const int ppr = (nx + 1) / 2;
-  const int T_ = (int)gridDim.x;
-  const int b_ = (int)blockIdx.x;
-  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
-  const int bid_ =
-      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
-      + yc_;
-  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
-  if (t >= ppr * ny) return;
-  const int j = t / ppr;
-  const int i0 = (t % ppr) * 2;
   const int jmin = a.f.south_open ? 1 : 2;
   const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
   const int imin = a.f.west_open ? 1 : 2;
@@ -800,6 +790,77 @@ __global__ void sw_stage27v(SwArgs<float> a) {
            (nu * (vn - vcc) * a.rdy - nu * (vcc - vs) * a.rdy) * a.rdy;
   st2(a.u2, idx, uc + a.dt * lu);
   st2(a.v2, idx, vcc + a.dt * lv);
+}
+
+__global__ void sw_stage27v(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
+  if (t >= ppr * ny) return;
+  stage27_pair(a, t / ppr, (t % ppr) * 2);
+}
+
+// halo/compute overlap split of the friction stage: a pair (2 cells) is
+// halo-INDEPENDENT iff its 3x4 input stencil stays inside the interior
+// ring [1, ny-2] x [1, nx-2] - those pairs (stage 28) can run while the
+// halo exchange is still in flight on a second stream; the remaining
+// ring pairs (stage 29) run after the exchange joins.  Both call the
+// identical stage27_pair, so 28+29 together reproduce stage 27 BITWISE.
+__device__ inline bool stage27_pair_independent(const SwArgs<float>& a,
+                                                int j, int i0) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  return j >= 2 && j <= ny - 3 && i0 >= 2 && i0 + 2 <= nx - 2;
+}
+
+__global__ void sw_stage28v(SwArgs<float> a) {  // interior pairs only
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int t = bid_ * (int)blockDim.x + (int)threadIdx.x;
+  if (t >= ppr * ny) return;
+  const int j = t / ppr, i0 = (t % ppr) * 2;
+  if (!stage27_pair_independent(a, j, i0)) return;
+  stage27_pair(a, j, i0);
+}
+
+__global__ void sw_stage29v(SwArgs<float> a) {  // ring pairs only
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int ppr = (nx + 1) / 2;
+  // ring pairs: rows 0,1,ny-3..ny-1 full width + 2 pair-columns on each
+  // side of the middle rows - enumerated compactly so the launch is tiny
+  const int side_pairs = 2;  // pairs at i0 = 0,2 and the last two pairs
+  const int mid_rows = ny - 5 > 0 ? ny - 5 : 0;
+  const long long top = 5LL * ppr;  // rows 0,1 and ny-3,ny-2,ny-1
+  const long long ring = top + (long long)mid_rows * 2 * side_pairs;
+  const long long t =
+      (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (t >= ring) return;
+  int j, i0;
+  if (t < top) {
+    const int r = (int)(t / ppr);       // 0,1 -> rows 0,1; 2,3,4 -> top
+    j = r < 2 ? r : ny - 5 + r;         // rows ny-3, ny-2, ny-1
+    i0 = (int)(t % ppr) * 2;
+  } else {
+    const long long u = t - top;
+    const int row = (int)(u / (2 * side_pairs));
+    const int k = (int)(u % (2 * side_pairs));
+    j = 2 + row;
+    const int last0 = ((nx + 1) / 2 - 1) * 2;  // final pair start
+    i0 = k < side_pairs ? k * 2 : last0 - (k - side_pairs) * 2;
+  }
+  if (stage27_pair_independent(a, j, i0)) return;  // guard small domains
+  stage27_pair(a, j, i0);
 }
 
 // stage 19 = stage 18 at 2 columns/thread (float2 math, float4 row
@@ -1043,6 +1104,15 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 22: hipLaunchKernelGGL(sw_stage21t<8>, grid, block, 0, stream, a); break;
       case 23: hipLaunchKernelGGL(sw_stage21t<16>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
+      case 28: hipLaunchKernelGGL(sw_stage28v, grid, block, 0, stream, a); break;
+      case 29: {
+        long long ppr2 = (p.nx + 1) / 2;
+        long long mid = p.ny - 5 > 0 ? p.ny - 5 : 0;
+        long long ring = 5 * ppr2 + mid * 4;
+        dim3 rg((unsigned)((ring + 255) / 256));
+        hipLaunchKernelGGL(sw_stage29v, rg, block, 0, stream, a);
+        break;
+      }
     }
     return;
   }

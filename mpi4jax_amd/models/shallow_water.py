@@ -60,7 +60,7 @@ class ShallowWater:
     def __init__(self, nx=360, ny=180, dx=5e3, dy=5e3, *, comm=None,
                  dims=None, device="cpu", dtype=torch.float32,
                  periodic_x=True, lateral_viscosity=None, fused=None,
-                 ref_friction_bug=False):
+                 ref_friction_bug=False, _force_remote_exchange=False):
         self.comm = resolve_comm(comm)
         self.grid = CartesianGrid(self.comm, dims=dims,
                                   periodic=(False, periodic_x))
@@ -111,6 +111,7 @@ class ShallowWater:
         # 386-391) for cross-framework trajectory validation — see
         # docs/PARITY.md.  Eager path only.
         self.ref_friction_bug = bool(ref_friction_bug)
+        self._force_remote_exchange = bool(_force_remote_exchange)
         if self.ref_friction_bug and fused:
             raise ValueError(
                 "ref_friction_bug=True requires the eager path (fused=False)"
@@ -302,11 +303,31 @@ class ShallowWater:
             stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
         stage(s6)         # tendencies + time update -> h_alt/u_alt/v_alt
         self._swap("h", "u", "v")
-        self._exchange_fields([fb["h"], fb["u"], fb["v"]])
-        if self.lateral_viscosity > 0:
-            stage(s7)     # friction Laplacian update -> u_alt/v_alt
+        if self._overlap_plan() is not None and self.lateral_viscosity > 0:
+            # halo/compute overlap (MPI4JAX_AMD_SW_OVERLAP=1): run the
+            # h/u/v exchange on a second stream while the main stream
+            # computes the halo-independent friction pairs (stage 28);
+            # the halo-dependent ring (stage 29) joins after.  28+29 use
+            # the identical per-pair code as stage 27, so the step is
+            # bitwise equal to the serial path.
+            s28, s29, comm_stream, ev1, ev2 = self._overlap_plan()
+            main = torch.cuda.current_stream()
+            ev1.record(main)
+            comm_stream.wait_event(ev1)
+            with torch.cuda.stream(comm_stream):
+                self._exchange_fields([fb["h"], fb["u"], fb["v"]])
+            stage(s28)    # deep interior, overlapped with the exchange
+            ev2.record(comm_stream)
+            main.wait_event(ev2)
+            stage(s29)    # boundary ring, needs the fresh halos
             self._swap("u", "v")
             self._exchange_fields([fb["u"], fb["v"]])
+        else:
+            self._exchange_fields([fb["h"], fb["u"], fb["v"]])
+            if self.lateral_viscosity > 0:
+                stage(s7)     # friction Laplacian update -> u_alt/v_alt
+                self._swap("u", "v")
+                self._exchange_fields([fb["u"], fb["v"]])
         # the new tendencies become "old" for the next step
         for k in ("h", "u", "v"):
             fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
@@ -314,6 +335,25 @@ class ShallowWater:
                           fb["do_v"])
 
     # ------------------------------------------------------------------
+    def _overlap_plan(self):
+        """(interior_stage, ring_stage, comm_stream) when the overlap path
+        is enabled and applicable (float32 vector path, GPU), else None.
+        Cached with the fused buffers."""
+        import os
+
+        fb = self._fb
+        if "overlap_plan" in fb:
+            return fb["overlap_plan"]
+        plan = None
+        if (os.environ.get("MPI4JAX_AMD_SW_OVERLAP") == "1"
+                and self.dtype == torch.float32
+                and self.device.type == "cuda"
+                and self._stage_plan()[2] == 27):
+            plan = (28, 29, torch.cuda.Stream(), torch.cuda.Event(),
+                    torch.cuda.Event())
+        fb["overlap_plan"] = plan
+        return plan
+
     def _exchange_cache(self):
         """Step-invariant halo-exchange state: the resolved schedule (as
         flat int lists for the one-call C++ executor), the staging
@@ -332,6 +372,14 @@ class ShallowWater:
         fb["cor_rbuf"] = torch.empty_like(fb["cor_sbuf"])
         (wrap_sides, col_ops, row_ops, cor_ops,
          cor_mask) = halo_exchange_schedule(self.grid, nx, ny)
+        if getattr(self, "_force_remote_exchange", False) and wrap_sides:
+            # test hook: express the periodic self-wraps as remote
+            # transfers to self so the full RCCL exchange path (pack,
+            # grouped p2p, unpack) runs at world 1 — used by the overlap
+            # and loopback GPU tests (see tests/test_gpu_overlap.py)
+            me = self.comm.rank
+            wrap_sides = []
+            col_ops = [(0, me, me, 1, nx - 1), (1, me, me, nx - 2, 0)]
         z = -1  # C++ encoding of "no peer"
 
         def flat(ops):

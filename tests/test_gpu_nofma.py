@@ -7,8 +7,9 @@ experiment (profiles/fma_experiment_r02.md, tools/fma_diag.py on MI355X)
 shows that attribution was WRONG in an informative way:
 
 * the ``_rccl_C_nofma`` build (identical kernels, ``-ffp-contract=off``)
-  produces trajectories BITWISE IDENTICAL to the normal build — FMA
-  contraction contributes nothing to the divergence;
+  differs from the normal build by only ~4e-14 after 30 f64 steps —
+  ulp-level, 6 orders below the fused-vs-eager gap: FMA contraction is
+  subdominant;
 * both fused and eager paths are bitwise deterministic across runs (no
   races);
 * the divergence appears on the very first step (u: 8e-11, v: 2e-9 at
@@ -75,18 +76,38 @@ def _run_traj(out_path, nofma):
     return torch.load(out_path)
 
 
-def test_fma_contraction_plays_no_role():
-    """The -ffp-contract=off build must reproduce the normal build
-    BITWISE over 30 fused f64 steps — proving the fused-vs-eager
-    divergence is evaluation-order rounding, not FMA."""
+def test_fma_contraction_is_subdominant():
+    """Measured on MI355X (tools/fma_cross.py): the -ffp-contract=off
+    build differs from the normal build by at most ~4e-14 after 30 fused
+    f64 steps (ulp-level contraction effects in u/v), while the
+    fused-vs-eager divergence at the same horizon is ~7e-8 — evaluation
+    -order rounding dominates FMA by ~6 orders of magnitude.  Pin both
+    the cross-build bound and the dominance ratio."""
     if not os.path.exists(os.path.join(REPO, "mpi4jax_amd",
                                        "_rccl_C_nofma.so")):
         pytest.skip("nofma variant not built (run setup.py --nofma)")
+    import mpi4jax_amd as m
+    from mpi4jax_amd.models import ShallowWater
+
     with tempfile.TemporaryDirectory() as td:
         a = _run_traj(os.path.join(td, "fma.pt"), nofma=False)
         b = _run_traj(os.path.join(td, "nofma.pt"), nofma=True)
-    for k in ("h", "u", "v"):
-        assert torch.equal(a[k], b[k]), k
+    cross = max((a[k] - b[k]).abs().max().item() for k in ("h", "u", "v"))
+    assert cross < 1e-12, cross  # observed 4.3e-14 at 30 steps
+
+    # same-horizon fused-vs-eager divergence for the dominance ratio
+    m.init()
+    torch.manual_seed(0)
+    eager = ShallowWater(nx=128, ny=96, device="cuda",
+                         dtype=torch.float64, fused=False)
+    s = eager.initial_conditions()
+    s = eager.step(s, first_step=True)
+    for _ in range(30):
+        s = eager.step(s)
+    torch.cuda.synchronize()
+    order_err = max((a[k] - getattr(s, k).cpu()).abs().max().item()
+                    for k in ("h", "u", "v"))
+    assert order_err > 100 * max(cross, 1e-16), (order_err, cross)
 
 
 def test_fused_divergence_is_deterministic_order_rounding():
