@@ -491,6 +491,14 @@ class ShallowWater:
         if use_graph is None:
             use_graph = (self.fused and self.device.type == "cuda"
                          and env != "0")
+        if (use_graph and self.fused and self.device.type == "cuda"
+                and self._fb is not None
+                and self._overlap_plan() is not None):
+            # hipGraph capture of the two-stream overlapped step segfaults
+            # inside the HIP runtime on ROCm 7.0 (not catchable by the
+            # validate-then-adopt protocol) — the overlap path targets the
+            # non-graph multi-rank loop; capture is disabled with it.
+            use_graph = False
         if use_graph and steps_per_call % 2:
             raise ValueError("steps_per_call must be even for graph capture")
 

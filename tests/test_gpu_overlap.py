@@ -69,9 +69,12 @@ def test_overlap_with_forced_remote_exchange():
         assert torch.equal(base[k], ov[k]), k
 
 
-def test_overlap_graph_capture():
-    """The two-stream overlapped step must survive hipGraph capture with
-    the validate-then-adopt protocol (bitwise replay check inside)."""
+def test_overlap_disables_graph_capture():
+    """hipGraph capture of the two-stream overlapped step segfaults in
+    the HIP runtime on ROCm 7.0 (observed: SIGSEGV inside capture, not a
+    catchable error — gpurun_out/ov logs).  make_stepper must therefore
+    fall back to the eager loop when the overlap path is active, and the
+    stepper must still advance correctly."""
     os.environ["MPI4JAX_AMD_SW_OVERLAP"] = "1"
     try:
         torch.manual_seed(0)
