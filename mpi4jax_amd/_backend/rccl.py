@@ -275,16 +275,25 @@ def send(x, dest, tag, comm):
         if dest == comm.rank:
             _self_queue(comm).append((tag, xc.clone()))  # stream-ordered copy
             return
-        _check_gpu_tag(tag, "send")
+        from . import envelope
+
+        if envelope.enabled() and comm.gloo_group is not None:
+            envelope.box_for(comm).post(dest, tag,
+                                        xc.numel() * xc.element_size())
+        else:
+            _check_gpu_tag(tag, "send")
         ext().send(_bytes(xc), dest, _handle(comm))
 
 
 def recv(template, source, tag, comm, status):
-    if source == ANY_SOURCE:
+    from . import envelope
+
+    env_on = envelope.enabled() and comm.gloo_group is not None
+    if source == ANY_SOURCE and not env_on:
         raise ValueError(
-            "recv: ANY_SOURCE is not supported on the RCCL backend — "
-            "RCCL has no message envelope; pass an explicit source rank "
-            "(shapes are static so the source is always known)"
+            "recv: ANY_SOURCE on the RCCL backend requires the envelope "
+            "plane (set MPI4JAX_AMD_GPU_ENVELOPE=1) — RCCL itself has no "
+            "message envelope; alternatively pass an explicit source rank"
         )
     with debug_timer("Recv", comm.rank, f"from {source}, tag {tag}"):
         out = torch.empty(tuple(template.shape), dtype=template.dtype,
@@ -298,6 +307,23 @@ def recv(template, source, tag, comm, status):
                     _fill_status(status, source, t, out)
                     return out
             raise RuntimeError("recv from self with no matching buffered send")
+        if env_on:
+            h = _handle(comm)
+
+            def recv_bytes(src, nbytes):
+                buf = torch.empty(nbytes, dtype=torch.uint8,
+                                  device=out.device)
+                ext().recv(buf, src, h)
+                return buf
+
+            s, t, data = envelope.box_for(comm).claim(source, tag,
+                                                      recv_bytes)
+            if data is not None:  # drained ahead of order into the stash
+                out.copy_(data.view(out.dtype).reshape(out.shape))
+            else:
+                ext().recv(_bytes(out), s, h)
+            _fill_status(status, s, t, out)
+            return out
         _check_gpu_tag(tag, "recv")
         ext().recv(_bytes(out), source, _handle(comm))
         _fill_status(status, source, tag, out)
