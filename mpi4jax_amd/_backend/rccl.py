@@ -77,7 +77,35 @@ def ext_is_loaded():
     return _EXT is not None
 
 
+# comm -> the HIP stream the previous collective was enqueued on.  The
+# reference's JAX ordered effect (utils.py:45-53 token threading) pinned
+# "program order == network order" against compiler reordering; the torch
+# analog of that hazard is a user enqueuing collectives on DIFFERENT
+# streams: per-stream they'd execute in submission race order, and a
+# reorder against another rank's sequence deadlocks RCCL.  The fence makes
+# each collective's stream wait on the previous collective's stream
+# (event bridge), so per-communicator program order == network order even
+# across streams — the ordered-effect guarantee, enforced in hardware.
+_ORDER_STREAMS = weakref.WeakKeyDictionary()
+
+
+def _order_fence(comm):
+    if torch.cuda.is_current_stream_capturing():
+        # an event bridge from a non-capture stream is illegal inside
+        # capture; the graph path pins its own stream discipline
+        # (models/shallow_water.make_stepper captures on ONE stream)
+        return
+    cur = torch.cuda.current_stream()
+    prev = _ORDER_STREAMS.get(comm)
+    if prev is not None and prev != cur:
+        ev = torch.cuda.Event()
+        ev.record(prev)
+        cur.wait_event(ev)
+    _ORDER_STREAMS[comm] = cur
+
+
 def _handle(comm):
+    _order_fence(comm)
     return comm.rccl_handle()
 
 

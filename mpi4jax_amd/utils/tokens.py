@@ -2,11 +2,15 @@
 
 The reference removed user-visible tokens in 0.8.0 and raises when one is
 passed (``/root/reference/mpi4jax/_src/utils.py:30-42``); ordering is instead
-enforced internally by a JAX ordered effect.  Here ordering is enforced by
-HIP **stream order**: every collective is enqueued on the current compute
-stream, so program order == stream order == network order per communicator,
-with no token plumbing needed.  We keep the ``token=NOTSET`` kwarg and raise
-the same way for drop-in compatibility.
+enforced internally by a JAX ordered effect (token threading,
+``utils.py:45-53``).  Here the same guarantee — program order == network
+order per communicator — is enforced in hardware: every collective is
+enqueued on the current compute stream (program order == stream order),
+and when consecutive collectives land on *different* HIP streams the
+backend bridges them with a stream event
+(``_backend/rccl._order_fence``), so not even multi-stream code can
+reorder a communicator's sequence against another rank's.  We keep the
+``token=NOTSET`` kwarg and raise the same way for drop-in compatibility.
 """
 
 

@@ -164,6 +164,53 @@ def test_sw_exchange_partial_schedule_rows_only():
     assert torch.equal(f, e)
 
 
+def test_envelope_stash_drain_device_order():
+    """Device side of the envelope plane (_backend/envelope.py): when a
+    recv wants the SECOND message in a peer's pipe, claim() must drain the
+    first into a device stash with real RCCL recvs in send order.  The
+    protocol itself is pinned at world 2 over gloo (tests/test_envelope.py);
+    here the data plane is real RCCL with peer = self, envelopes
+    pre-queued (the gloo leg carries no data, so skipping it changes
+    nothing on the device side)."""
+    from mpi4jax_amd._backend.envelope import EnvelopeBox
+    from mpi4jax_amd.parallel.comm import get_default_comm
+
+    ext = _ext()
+    h = _handle()
+    box = EnvelopeBox(get_default_comm())
+    a = torch.full((16,), 7, dtype=torch.uint8, device="cuda")
+    b = torch.arange(32, dtype=torch.uint8, device="cuda")
+
+    drained = []
+
+    def recv_bytes(src, nbytes):
+        buf = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
+        ext.recv(buf, src, h)
+        drained.append((src, nbytes))
+        return buf
+
+    # all p2p with self must share one group (RCCL self-matching)
+    ext.group_start()
+    ext.send(a, 0, h)            # tag-7 message, first in the pipe
+    ext.send(b, 0, h)            # tag-9 message, wanted first by the recv
+    box.queue.append([0, 7, 16])
+    box.queue.append([0, 9, 32])
+    s, t, data = box.claim(0, 9, recv_bytes)
+    assert (s, t, data) == (0, 9, None)
+    assert drained == [(0, 16)], "tag-7 must drain (in send order) first"
+    out = torch.empty(32, dtype=torch.uint8, device="cuda")
+    ext.recv(out, 0, h)
+    ext.group_end()
+    torch.cuda.synchronize()
+
+    assert torch.equal(out, b)
+    s, t, data = box.claim(0, 7, recv_bytes)  # satisfied from the stash
+    assert (s, t) == (0, 7) and data is not None
+    torch.cuda.synchronize()
+    assert torch.equal(data, a)
+    assert box.queue == [] and box.stash == []
+
+
 def test_scan_chain_sequence_self():
     """The scan ring's per-rank native sequence (bridge.cpp:446-468):
     recv the running prefix, combine on the CDNA4 kernel, send onward —
