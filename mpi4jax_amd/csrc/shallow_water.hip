@@ -234,22 +234,25 @@ __device__ inline T ke_at(const SwArgs<T>& a, int j, int i) {
 }
 
 // stage 8 = stage 1 + stage 6 fused: derived fields evaluated in-register,
-// no fe/fn/q/ke array traffic at all.
+// no fe/fn/q/ke array traffic at all.  The math lives in stage8_math so
+// the fused update+friction stage (30) can evaluate a cell's update
+// without storing it (same expressions, same order -> same rounding).
 template <typename T>
-__device__ inline void stage8_cell(const SwArgs<T>& a, int j, int i) {
+struct CellUpd {
+  T dnh, dnu, dnv, hh, uu, vv;
+};
+
+template <typename T>
+__device__ inline CellUpd<T> stage8_math(const SwArgs<T>& a, int j, int i) {
+  // caller guarantees an interior cell (1 <= j <= ny-2, 1 <= i <= nx-2)
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int idx = j * nx + i;
   T h_ = a.h[idx], u_ = a.u[idx], v_ = a.v[idx];
-  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
-    a.h2[idx] = h_;
-    a.u2[idx] = u_;
-    a.v2[idx] = v_;
-    return;
-  }
+  CellUpd<T> r;
 
   T dnh = -(fe_at(a, j, i) - fe_at(a, j, i - 1)) / a.dx -
           (fn_at(a, j, i) - fn_at(a, j - 1, i)) / a.dy;
-  a.dnh[idx] = dnh;
+  r.dnh = dnh;
 
   T qc = q_at(a, j, i), qs = q_at(a, j - 1, i), qw = q_at(a, j, i - 1);
   T fnc = fn_at(a, j, i), fne = fn_at(a, j, i + 1);
@@ -258,7 +261,7 @@ __device__ inline void stage8_cell(const SwArgs<T>& a, int j, int i) {
           T(0.5) * (qc * T(0.5) * (fnc + fne) +
                     qs * T(0.5) * (fns + fnse));
   dnu -= (ke_at(a, j, i + 1) - ke_at(a, j, i)) / a.dx;
-  a.dnu[idx] = dnu;
+  r.dnu = dnu;
 
   T fec = fe_at(a, j, i), fen = fe_at(a, j + 1, i);
   T few = fe_at(a, j, i - 1), fenw = fe_at(a, j + 1, i - 1);
@@ -266,15 +269,35 @@ __device__ inline void stage8_cell(const SwArgs<T>& a, int j, int i) {
           T(0.5) * (qc * T(0.5) * (fec + fen) +
                     qw * T(0.5) * (few + fenw));
   dnv -= (ke_at(a, j + 1, i) - ke_at(a, j, i)) / a.dy;
-  a.dnv[idx] = dnv;
+  r.dnv = dnv;
 
   T uu = u_ + a.dt * (a.ab_a * dnu + a.ab_b * a.dou[idx]);
   T vv = v_ + a.dt * (a.ab_a * dnv + a.ab_b * a.dov[idx]);
-  a.h2[idx] = h_ + a.dt * (a.ab_a * dnh + a.ab_b * a.doh[idx]);
+  r.hh = h_ + a.dt * (a.ab_a * dnh + a.ab_b * a.doh[idx]);
   if (a.f.east_wall && i == nx - 2) uu = T(0);
   if (a.f.north_wall && j == ny - 2) vv = T(0);
-  a.u2[idx] = uu;
-  a.v2[idx] = vv;
+  r.uu = uu;
+  r.vv = vv;
+  return r;
+}
+
+template <typename T>
+__device__ inline void stage8_cell(const SwArgs<T>& a, int j, int i) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int idx = j * nx + i;
+  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+    a.h2[idx] = a.h[idx];
+    a.u2[idx] = a.u[idx];
+    a.v2[idx] = a.v[idx];
+    return;
+  }
+  CellUpd<T> r = stage8_math(a, j, i);
+  a.dnh[idx] = r.dnh;
+  a.dnu[idx] = r.dnu;
+  a.dnv[idx] = r.dnv;
+  a.h2[idx] = r.hh;
+  a.u2[idx] = r.uu;
+  a.v2[idx] = r.vv;
 }
 
 template <typename T>
@@ -891,21 +914,30 @@ __device__ inline void st2s(float* p, long long off, vf2 v) {
   }
 }
 
-template <bool NT>
-__device__ inline void stage19_cells(const SwArgs<float>& a, int j,
-                                     int i0) {
+// the pair-fast predicate of the 2-column update kernels: matches
+// stage19_cells' fast branch exactly (the ld4 at idn reads offsets 0..3,
+// so the pair must keep i0+3 in-row)
+__device__ inline bool stage19_pair_fast(const SwArgs<float>& a, int j,
+                                         int i0) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int jmin = a.f.south_open ? 1 : 2;
   const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
   const int imin = a.f.west_open ? 1 : 2;
   const int imax = (a.f.east_open && !a.f.east_wall) ? nx - 1 : nx - 2;
-  // the ld4 at idn reads offsets 0..3; keep it in-row (i0+3 <= nx-1)
-  const bool fast = j >= jmin && j <= jmax && i0 >= imin &&
-                    i0 + 2 <= imax && i0 + 3 < nx;
-  if (!fast) {
-    for (int c = 0; c < 2 && i0 + c < nx; ++c) stage8_cell(a, j, i0 + c);
-    return;
-  }
+  return j >= jmin && j <= jmax && i0 >= imin && i0 + 2 <= imax &&
+         i0 + 3 < nx;
+}
+
+struct Upd2 {
+  vf2 dnh, dnu, dnv, hh, uu, vv;
+};
+
+// fast-region update math for one 2-column pack (the stage19 body);
+// caller guarantees stage19_pair_fast(a, j, i0)
+template <bool NT>
+__device__ inline Upd2 stage19_math(const SwArgs<float>& a, int j,
+                                    int i0) {
+  const int nx = (int)a.nx;
   const long long idx = (long long)j * nx + i0;
   const long long idn = idx - nx, idp = idx + nx;
 
@@ -967,12 +999,32 @@ __device__ inline void stage19_cells(const SwArgs<float>& a, int j,
                     q_w * 0.5f * (fe_w + fe_nw));
   dnv -= (ke_n - ke_c) * rdy;
 
-  st2s<NT>(a.dnh, idx, dnh);
-  st2s<NT>(a.dnu, idx, dnu);
-  st2s<NT>(a.dnv, idx, dnv);
-  st2s<NT>(a.h2, idx, H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh));
-  st2s<NT>(a.u2, idx, U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou));
-  st2s<NT>(a.v2, idx, V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov));
+  Upd2 r;
+  r.dnh = dnh;
+  r.dnu = dnu;
+  r.dnv = dnv;
+  r.hh = H00 + a.dt * (a.ab_a * dnh + a.ab_b * doh);
+  r.uu = U00 + a.dt * (a.ab_a * dnu + a.ab_b * dou);
+  r.vv = V00 + a.dt * (a.ab_a * dnv + a.ab_b * dov);
+  return r;
+}
+
+template <bool NT>
+__device__ inline void stage19_cells(const SwArgs<float>& a, int j,
+                                     int i0) {
+  const int nx = (int)a.nx;
+  if (!stage19_pair_fast(a, j, i0)) {
+    for (int c = 0; c < 2 && i0 + c < nx; ++c) stage8_cell(a, j, i0 + c);
+    return;
+  }
+  const long long idx = (long long)j * nx + i0;
+  Upd2 r = stage19_math<NT>(a, j, i0);
+  st2s<NT>(a.dnh, idx, r.dnh);
+  st2s<NT>(a.dnu, idx, r.dnu);
+  st2s<NT>(a.dnv, idx, r.dnv);
+  st2s<NT>(a.h2, idx, r.hh);
+  st2s<NT>(a.u2, idx, r.uu);
+  st2s<NT>(a.v2, idx, r.vv);
 }
 
 template <bool NT>
@@ -1016,6 +1068,245 @@ __global__ void sw_stage21t(SwArgs<float> a) {
   if (j >= ny || i0 >= nx) return;
   stage19_cells<false>(a, j, i0);
 }
+
+// ------------------------------------------------ stage 30: fused step
+// Update + friction in one pass (world-1 / fully-local halos only).
+// The two-kernel pipeline writes the post-update u'/v' to HBM, wrap-
+// exchanges their halos, and reads them back for the friction Laplacian:
+// 4 of the step's 16 field passes plus one whole exchange exist only to
+// ferry that intermediate.  Stage 30 keeps u'/v' in LDS instead:
+// blocks compute the stage-19 update for a (TJ+2)-row tile into LDS
+// (vertical recompute replaces the cross-block dependency) and apply
+// the friction to the TJ middle rows from LDS.
+//
+// Numerics: same formulas in the same order as the two-kernel path
+// (stage19_math / stage8_math / stage27's rdx form), with the wrap
+// exchange synthesized by uvprime_cell.  NOT bitwise: the compiler
+// contracts the shared expression trees to FMA differently per inlining
+// site (~1 ulp/step, tolerance-tested in tests/test_gpu_ops.py::
+// test_stage30_matches_two_kernel_path).
+
+// The boundary ring's friction needs the post-update u'/v' — instead of
+// recomputing them per stencil point (measured 26 us/step), the fast
+// kernel and the update-ring kernel STORE the ring strip's u'/v' into
+// the fe/fn scratch fields (unused on the fused path), and the
+// friction-ring kernel just reads them.  Ring strip = the cleanup
+// enumeration: rows 0..3 / ny-4..ny-1, cols 0..5 / nx-6..nx-1 — covers
+// every not-s30_fast_out cell, its 4 stencil neighbors, and the wrap
+// sources (rows/cols 1, ny-2, nx-2).
+__device__ inline bool s30_in_ring(const SwArgs<float>& a, int jj,
+                                   int i0) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  return jj <= 3 || jj >= ny - 4 || i0 <= 5 || i0 >= nx - 7;
+}
+
+// what the ring friction reads at (jj, ii): the mid-step exchange's
+// value — open-edge halos wrap to their (world-1) periodic source,
+// closed-edge halos pass the pre-update field through, interior cells
+// read the stored u'/v' strip.  World>1 never runs stage 30 (remote
+// halo values cannot be synthesized locally).
+__device__ inline void s30_read_uv(const SwArgs<float>& a, int jj, int ii,
+                                   float* uu, float* vv) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  int js = jj, is = ii;
+  if (js == 0 && a.f.south_open) js = ny - 2;
+  else if (js == ny - 1 && a.f.north_open) js = 1;
+  if (is == 0 && a.f.west_open) is = nx - 2;
+  else if (is == nx - 1 && a.f.east_open) is = 1;
+  const long long idx = (long long)js * nx + is;
+  if (js < 1 || js > ny - 2 || is < 1 || is > nx - 2) {
+    *uu = a.u[idx];  // closed-axis halo: pass-through
+    *vv = a.v[idx];
+  } else {
+    *uu = a.fe[idx];  // u'/v' strip staged by stage30v / ringA
+    *vv = a.fn[idx];
+  }
+}
+
+// Tiling: TJ=14 output rows x 60 output columns per 256-thread block,
+// tile (TJ+2) x 64: the LDS fill is exactly 2 full 256-thread rounds
+// and redundant update compute is 16/14 x 64/60 = 1.22x (vs 3x for a
+// 1-row tile).  The fast kernel is PURE vector code — every scalar/
+// boundary fallback lives in the ring kernels below, keeping the fast
+// kernel at stage 19's VGPR budget (70 -> 7 waves/SIMD).  Store
+// partition between the kernels is exact and index-only: dn*/h2 by
+// stage19_pair_fast, u2/v2 by s30_fast_out.
+constexpr int kTJ30 = 14;    // output rows per block
+constexpr int kPC30 = 30;    // output pairs (2 cols) per row
+constexpr int kTR30 = kTJ30 + 2;           // tile rows
+constexpr int kTP30 = kPC30 + 2;           // tile pairs per row
+constexpr int kFill30 = kTR30 * kTP30;     // LDS fill tasks
+
+// friction output (j, pair i0) is servable from the fast kernel iff the
+// nine (row, pair) LDS inputs it needs are all stage19-fast and the pair
+// is stage27-fast; folds to pure bounds:
+__device__ inline bool s30_fast_out(const SwArgs<float>& a, int j,
+                                    int i0) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int jmin = a.f.south_open ? 1 : 2;
+  const int jmax = (a.f.north_open && !a.f.north_wall) ? ny - 2 : ny - 3;
+  const int imin = a.f.west_open ? 1 : 2;
+  const int imax = (a.f.east_open && !a.f.east_wall) ? nx - 1 : nx - 2;
+  return j - 1 >= jmin && j + 1 <= jmax && i0 - 2 >= imin &&
+         i0 + 4 <= imax && i0 + 5 < nx;
+}
+
+__global__ void sw_stage30v(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int bpc = (nx + 2 * kPC30 - 1) / (2 * kPC30);  // blocks per band
+  const int T_ = (int)gridDim.x;
+  const int b_ = (int)blockIdx.x;
+  const int q8_ = T_ / 8, r8_ = T_ % 8, xc_ = b_ % 8, yc_ = b_ / 8;
+  const int bid_ =
+      (xc_ < r8_ ? xc_ * (q8_ + 1) : r8_ * (q8_ + 1) + (xc_ - r8_) * q8_)
+      + yc_;
+  const int jb = (bid_ / bpc) * kTJ30;       // first owned output row
+  const int cb = (bid_ % bpc) * 2 * kPC30;   // first owned output column
+  const int tid = (int)threadIdx.x;
+
+  __shared__ float su[kTR30][2 * kTP30], sv[kTR30][2 * kTP30];
+  for (int t = tid; t < kFill30; t += 256) {
+    const int lr = t / kTP30, lp = t % kTP30;
+    const int rr = jb - 1 + lr;
+    const int i0 = cb - 2 + 2 * lp;
+    if (rr < 0 || !stage19_pair_fast(a, rr, i0)) continue;
+    Upd2 w = stage19_math<false>(a, rr, i0);
+    su[lr][2 * lp] = w.uu.x;
+    su[lr][2 * lp + 1] = w.uu.y;
+    sv[lr][2 * lp] = w.vv.x;
+    sv[lr][2 * lp + 1] = w.vv.y;
+    if (lr >= 1 && lr <= kTJ30 && lp >= 1 && lp <= kPC30) {
+      const long long idx = (long long)rr * nx + i0;  // owned pair
+      st2(a.dnh, idx, w.dnh);
+      st2(a.dnu, idx, w.dnu);
+      st2(a.dnv, idx, w.dnv);
+      st2(a.h2, idx, w.hh);
+      if (s30_in_ring(a, rr, i0)) {  // stage u'/v' for the ring friction
+        st2(a.fe, idx, w.uu);
+        st2(a.fn, idx, w.vv);
+      }
+    }
+  }
+  __syncthreads();
+
+  for (int t = tid; t < kTJ30 * kPC30; t += 256) {
+    const int ro = t / kPC30, po = t % kPC30;
+    const int j = jb + ro;
+    const int i0 = cb + 2 * po;
+    if (!s30_fast_out(a, j, i0)) continue;  // ring kernels own this pair
+    const int lr = ro + 1, x = 2 * po + 2;
+    const long long idx = (long long)j * nx + i0;
+    // stage27_pair math, inputs from LDS
+    vf2 uw = {su[lr][x - 1], su[lr][x]}, uc = {su[lr][x], su[lr][x + 1]},
+        ue = {su[lr][x + 1], su[lr][x + 2]};
+    vf2 un = {su[lr + 1][x], su[lr + 1][x + 1]},
+        us = {su[lr - 1][x], su[lr - 1][x + 1]};
+    vf2 vw = {sv[lr][x - 1], sv[lr][x]}, vcc = {sv[lr][x], sv[lr][x + 1]},
+        ve = {sv[lr][x + 1], sv[lr][x + 2]};
+    vf2 vn = {sv[lr + 1][x], sv[lr + 1][x + 1]},
+        vs = {sv[lr - 1][x], sv[lr - 1][x + 1]};
+    const float nu = a.nu;
+    vf2 lu = (nu * (ue - uc) * a.rdx - nu * (uc - uw) * a.rdx) * a.rdx +
+             (nu * (un - uc) * a.rdy - nu * (uc - us) * a.rdy) * a.rdy;
+    vf2 lv = (nu * (ve - vcc) * a.rdx - nu * (vcc - vw) * a.rdx) * a.rdx +
+             (nu * (vn - vcc) * a.rdy - nu * (vcc - vs) * a.rdy) * a.rdy;
+    st2(a.u2, idx, uc + a.dt * lu);
+    st2(a.v2, idx, vcc + a.dt * lv);
+  }
+}
+
+// Ring kernels: every output cell the fast kernel skips — a ~3-wide
+// perimeter.  Cells are enumerated as a compact ring (8 full edge rows +
+// 12 edge columns of each middle row); the predicates, not the
+// enumeration, define ownership, so over-coverage on tiny domains just
+// re-writes identical values.  ringA does the update outputs (and
+// stages u'/v' into fe/fn for the not-stage19-fast cells); ringB, a
+// separate kernel so the strip is complete, does the friction from the
+// staged strip.
+__device__ inline bool s30_ring_cell(const SwArgs<float>& a, long long t,
+                                     int* jo, int* io) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  const int edge_rows = ny < 8 ? ny : 8;
+  const int edge_cols = nx < 12 ? nx : 12;
+  const long long top = (long long)edge_rows * nx;
+  const int mid = ny > 8 ? ny - 8 : 0;
+  if (t >= top + (long long)mid * edge_cols) return false;
+  if (t < top) {
+    const int r = (int)(t / nx);
+    *jo = r < 4 ? r : ny - 8 + r;
+    *io = (int)(t % nx);
+  } else {
+    const long long u = t - top;
+    *jo = 4 + (int)(u / edge_cols);
+    const int c = (int)(u % edge_cols);
+    *io = c < 6 ? c : nx - 12 + c;
+  }
+  return true;
+}
+
+__global__ void sw_stage30_ringA(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  int j, i;
+  if (!s30_ring_cell(a, (long long)blockIdx.x * blockDim.x + threadIdx.x,
+                     &j, &i))
+    return;
+  const long long idx = (long long)j * nx + i;
+  if (stage19_pair_fast(a, j, i & ~1)) return;  // fast kernel owns it
+  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+    a.h2[idx] = a.h[idx];
+    return;
+  }
+  CellUpd<float> r8 = stage8_math(a, j, i);
+  a.dnh[idx] = r8.dnh;
+  a.dnu[idx] = r8.dnu;
+  a.dnv[idx] = r8.dnv;
+  a.h2[idx] = r8.hh;
+  a.fe[idx] = r8.uu;  // complete the u'/v' strip for ringB
+  a.fn[idx] = r8.vv;
+}
+
+__global__ void sw_stage30_ringB(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  int j, i;
+  if (!s30_ring_cell(a, (long long)blockIdx.x * blockDim.x + threadIdx.x,
+                     &j, &i))
+    return;
+  const long long idx = (long long)j * nx + i;
+  if (s30_fast_out(a, j, i & ~1)) return;  // fast kernel owns it
+  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+    a.u2[idx] = a.u[idx];  // pass-through; open halos are overwritten
+    a.v2[idx] = a.v[idx];  // by the end-of-step wrap refresh
+    return;
+  }
+  // stage7_cell math on the staged u'/v' strip
+  float uc, vc, un, vn, us, vs, ue, ve, uw, vw;
+  s30_read_uv(a, j, i, &uc, &vc);
+  s30_read_uv(a, j + 1, i, &un, &vn);
+  s30_read_uv(a, j - 1, i, &us, &vs);
+  s30_read_uv(a, j, i + 1, &ue, &ve);
+  s30_read_uv(a, j, i - 1, &uw, &vw);
+  auto gu = [&](float c, float e, int ii) -> float {
+    if (ii > nx - 2 || (ii == 0 && !a.f.west_open)) return 0.f;
+    if (a.f.east_wall && ii == nx - 2) return 0.f;
+    return a.nu * (e - c) / a.dx;
+  };
+  auto gv = [&](float c, float n, int jj) -> float {
+    if (jj > ny - 2 || (jj == 0 && !a.f.south_open)) return 0.f;
+    if (a.f.north_wall && jj == ny - 2) return 0.f;
+    return a.nu * (n - c) / a.dy;
+  };
+  const float lu = (gu(uc, ue, i) - gu(uw, uc, i - 1)) / a.dx +
+                   (gv(uc, un, j) - gv(us, uc, j - 1)) / a.dy;
+  const float lv = (gu(vc, ve, i) - gu(vw, vc, i - 1)) / a.dx +
+                   (gv(vc, vn, j) - gv(vs, vc, j - 1)) / a.dy;
+  float uu2 = uc + a.dt * lu;
+  float vv2 = vc + a.dt * lv;
+  if (a.f.east_wall && i == nx - 2) uu2 = 0.f;
+  if (a.f.north_wall && j == ny - 2) vv2 = 0.f;
+  a.u2[idx] = uu2;
+  a.v2[idx] = vv2;
+}
+
 
 int sw_grid(long long n) {
   long long blocks = (n + kBlock - 1) / kBlock;
@@ -1104,6 +1395,19 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 22: hipLaunchKernelGGL(sw_stage21t<8>, grid, block, 0, stream, a); break;
       case 23: hipLaunchKernelGGL(sw_stage21t<16>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
+      case 30: {
+        long long bpc = (p.nx + 2 * kPC30 - 1) / (2 * kPC30);
+        long long bands = (p.ny + kTJ30 - 1) / kTJ30;
+        dim3 g30((unsigned)(bpc * bands));
+        hipLaunchKernelGGL(sw_stage30v, g30, block, 0, stream, a);
+        long long er = p.ny < 8 ? p.ny : 8, ec = p.nx < 12 ? p.nx : 12;
+        long long ring = er * p.nx +
+                         (p.ny > 8 ? (p.ny - 8) * ec : 0);
+        dim3 gcl((unsigned)((ring + 255) / 256));
+        hipLaunchKernelGGL(sw_stage30_ringA, gcl, block, 0, stream, a);
+        hipLaunchKernelGGL(sw_stage30_ringB, gcl, block, 0, stream, a);
+        break;
+      }
       case 28: hipLaunchKernelGGL(sw_stage28v, grid, block, 0, stream, a); break;
       case 29: {
         long long ppr2 = (p.nx + 1) / 2;

@@ -264,6 +264,19 @@ class ShallowWater:
             if tile in ("4", "8", "16"):
                 # band-tiled tendency kernel: TJ rows per 256-thread block
                 return None, {"4": 21, "8": 22, "16": 23}[tile], 27
+            if (self.lateral_viscosity > 0
+                    and self.grid.nproc_y * self.grid.nproc_x == 1
+                    and not self._force_remote_exchange
+                    and not os.environ.get("MPI4JAX_AMD_SW_OVERLAP")
+                    and os.environ.get("MPI4JAX_AMD_SW_NOFUSE") != "1"):
+                # single-rank fully-local halos: update+friction fused
+                # (stage 30 fast kernel + boundary cleanup) — drops the
+                # u'/v' intermediate round trip (4 of 16 field passes)
+                # and the mid-step exchange; equivalent to the two-kernel
+                # path to FMA-contraction rounding (tests/test_gpu_ops.py
+                # ::test_stage30_matches_two_kernel_path).  s7=None marks
+                # the fused-friction step shape.
+                return None, 30, None
             # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
             # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
             return None, 19, 27
@@ -301,6 +314,17 @@ class ShallowWater:
 
         if s1 is not None:
             stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
+        if s6 == 30:
+            # fused update+friction: u'/v' never round-trip through HBM
+            # and the mid-step exchange disappears; one wrap refresh of
+            # the finals closes the step
+            stage(s6)
+            self._swap("h", "u", "v")
+            self._exchange_fields([fb["h"], fb["u"], fb["v"]])
+            for k in ("h", "u", "v"):
+                fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
+            return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"],
+                              fb["do_u"], fb["do_v"])
         stage(s6)         # tendencies + time update -> h_alt/u_alt/v_alt
         self._swap("h", "u", "v")
         if self._overlap_plan() is not None and self.lateral_viscosity > 0:

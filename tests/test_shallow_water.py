@@ -199,9 +199,18 @@ def test_stage_plan_selection(monkeypatch):
     sw = ShallowWater(nx=12, ny=6, device="cpu")
     for var in ("MPI4JAX_AMD_SW_MERGED", "MPI4JAX_AMD_SW_NOVEC",
                 "MPI4JAX_AMD_SW_TWOPASS", "MPI4JAX_AMD_SW_4COL",
-                "MPI4JAX_AMD_SW_NT"):
+                "MPI4JAX_AMD_SW_NT", "MPI4JAX_AMD_SW_NOFUSE"):
         monkeypatch.delenv(var, raising=False)
+    # single-rank fully-local halos: the fused update+friction kernel
+    assert sw._stage_plan() == (None, 30, None)
+    monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", "1")
     assert sw._stage_plan() == (None, 19, 27)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_NOFUSE")
+    # remote halos (forced) disqualify stage 30
+    sw_r = ShallowWater(nx=12, ny=6, device="cpu",
+                        _force_remote_exchange=True)
+    assert sw_r._stage_plan() == (None, 19, 27)
+    monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", "1")
     monkeypatch.setenv("MPI4JAX_AMD_SW_4COL", "1")
     assert sw._stage_plan() == (None, 18, 17)
     monkeypatch.delenv("MPI4JAX_AMD_SW_4COL")

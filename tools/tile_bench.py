@@ -18,6 +18,9 @@ from mpi4jax_amd.models import ShallowWater  # noqa: E402
 
 
 def traj(tile, steps=10):
+    # compare against the two-kernel stage-19 path (stage 30 fuses the
+    # friction and is compared separately in tests/test_gpu_ops.py)
+    os.environ["MPI4JAX_AMD_SW_NOFUSE"] = "1"
     if tile:
         os.environ["MPI4JAX_AMD_SW_TILE"] = tile
     else:
@@ -33,6 +36,7 @@ def traj(tile, steps=10):
 
 
 def bench(tile, nx=3600, ny=1800, steps=300, warmup=50):
+    os.environ["MPI4JAX_AMD_SW_NOFUSE"] = "1"
     if tile:
         os.environ["MPI4JAX_AMD_SW_TILE"] = tile
     else:
