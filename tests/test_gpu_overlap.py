@@ -51,7 +51,10 @@ def _traj(monkeypatch_env, steps=20, force_remote=False):
 
 
 def test_overlap_bitwise_equal_serial():
-    _, base = _traj({"MPI4JAX_AMD_SW_OVERLAP": None})
+    # serial baseline = the two-kernel 19+27 pipeline (28+29 reuse its
+    # pair code); stage 30, the world-1 default, fuses differently
+    _, base = _traj({"MPI4JAX_AMD_SW_OVERLAP": None,
+                     "MPI4JAX_AMD_SW_NOFUSE": "1"})
     sw, ov = _traj({"MPI4JAX_AMD_SW_OVERLAP": "1"})
     assert sw._overlap_plan() is not None, "overlap path not active"
     for k in ("h", "u", "v"):
@@ -61,7 +64,8 @@ def test_overlap_bitwise_equal_serial():
 def test_overlap_with_forced_remote_exchange():
     """Comm stream carries real RCCL self-transfers (pack, grouped p2p,
     unpack) while stage 28 runs on the main stream."""
-    _, base = _traj({"MPI4JAX_AMD_SW_OVERLAP": None})
+    _, base = _traj({"MPI4JAX_AMD_SW_OVERLAP": None,
+                     "MPI4JAX_AMD_SW_NOFUSE": "1"})
     sw, ov = _traj({"MPI4JAX_AMD_SW_OVERLAP": "1"}, force_remote=True)
     assert sw._overlap_plan() is not None
     assert sw._exchange_cache()["comm_id"] != -1, "RCCL path not used"

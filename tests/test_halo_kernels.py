@@ -316,6 +316,9 @@ def test_nt_stage_variant_matches(monkeypatch):
     from mpi4jax_amd.models import ShallowWater
 
     def run(nt):
+        # compare against stage 19 (the NT kernel is its cache-policy
+        # twin) — not the fused stage 30 world-1 default
+        monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", "1")
         if nt:
             monkeypatch.setenv("MPI4JAX_AMD_SW_NT", "1")
         else:
