@@ -54,6 +54,7 @@ def ext():
         else:
             from .. import _rccl_C  # type: ignore
 
+        _check_build_compat(_rccl_C)
         _EXT = _rccl_C
         _EXT.set_logging(get_logging())
         # opt-in fail-fast watchdog: a wedged p2p (mismatched send/recv,
@@ -71,6 +72,56 @@ def ext():
             f"(original error: {e})"
         )
         raise _EXT_ERR
+
+
+def _check_build_compat(ext_mod):
+    """Runtime-vs-build compatibility guard (the reference's MPI ABI
+    check, ``xla_bridge/__init__.py:23-89``, re-expressed for the pair
+    that matters here: the torch C++ ABI the extension was compiled
+    against, and the RCCL header vs the librccl actually loaded).
+
+    torch major.minor mismatch is an error (the C++ ABI is not stable
+    across minor releases; symptoms otherwise range from import errors
+    to silent corruption); RCCL header-vs-runtime skew is a warning
+    (the NCCL API is stable within a major version).  Bypass with
+    ``MPI4JAX_AMD_SKIP_ABI_CHECK=1`` — same escape hatch semantics as
+    the reference's ``MPI4JAX_SKIP_ABI_CHECK``.
+    """
+    import os
+    import warnings
+
+    if os.environ.get("MPI4JAX_AMD_SKIP_ABI_CHECK", "") == "1":
+        return
+    try:
+        built = dict(ext_mod.build_info())
+    except AttributeError:  # older .so without build_info
+        return
+    run_torch = torch.__version__.split("+")[0]
+    b = built["torch"].split(".")[:2]
+    r = run_torch.split(".")[:2]
+    if b != r:
+        raise ImportError(
+            f"mpi4jax_amd native extension was built against torch "
+            f"{built['torch']} but torch {run_torch} is running — the "
+            "torch C++ ABI is not stable across versions; rebuild with "
+            "`python setup.py build_ext --inplace` (or set "
+            "MPI4JAX_AMD_SKIP_ABI_CHECK=1 to proceed at your own risk)"
+        )
+    rt = dict(ext_mod.version_info()).get("rccl", 0)
+    bh = built.get("rccl_header", 0)
+    if rt and bh and rt // 10000 != bh // 10000:
+        raise ImportError(
+            f"RCCL major-version mismatch: extension built against "
+            f"header {bh} but librccl {rt} is loaded (set "
+            "MPI4JAX_AMD_SKIP_ABI_CHECK=1 to override)"
+        )
+    if rt and bh and rt != bh:
+        warnings.warn(
+            f"mpi4jax_amd: RCCL header/runtime skew (built {bh}, "
+            f"running {rt}) — API-compatible within a major version, "
+            "continuing",
+            stacklevel=3,
+        )
 
 
 def ext_is_loaded():

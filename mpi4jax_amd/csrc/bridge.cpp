@@ -14,6 +14,7 @@
 // into our own registry instead of raw MPI handles.
 
 #include <torch/extension.h>
+#include <torch/version.h>
 
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
@@ -333,6 +334,21 @@ py::dict version_info() {
   int hip_ver = 0;
   (void)hipRuntimeGetVersion(&hip_ver);
   d["hip_runtime"] = hip_ver;
+  return d;
+}
+
+// what this .so was COMPILED against — the runtime side (version_info)
+// is compared against this at import (the reference's runtime-vs-build
+// MPI ABI check, xla_bridge/__init__.py:23-89, re-expressed for the
+// torch-C++-ABI + RCCL pair that matters here)
+py::dict build_info() {
+  py::dict d;
+  d["torch"] = std::to_string(TORCH_VERSION_MAJOR) + "." +
+               std::to_string(TORCH_VERSION_MINOR) + "." +
+               std::to_string(TORCH_VERSION_PATCH);
+  d["rccl_header"] =
+      NCCL_MAJOR * 10000 + NCCL_MINOR * 100 + NCCL_PATCH;
+  d["glibcxx_use_cxx11_abi"] = (int)_GLIBCXX_USE_CXX11_ABI;
   return d;
 }
 
@@ -927,6 +943,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("get_watchdog", &get_watchdog);
   m.def("debug_wedge_stream", &debug_wedge_stream);
   m.def("version_info", &version_info);
+  m.def("build_info", &build_info);
   m.def("allreduce", &allreduce);
   m.def("reduce", &reduce);
   m.def("allgather", &allgather);

@@ -65,3 +65,42 @@ def test_string_op_names():
     assert torch.equal(m.scan(x, "MAX"), x)
     with pytest.raises(TypeError):
         m.allreduce(x, "not-an-op")
+
+
+# ------------------------------------------- build/runtime compat guard
+
+def test_build_compat_guard(monkeypatch):
+    """Runtime-vs-build guard (reference MPI ABI check analog): torch
+    major.minor mismatch and RCCL major mismatch raise; minor RCCL skew
+    only warns; env bypass works."""
+    import warnings
+
+    import torch
+
+    from mpi4jax_amd._backend.rccl import _check_build_compat
+
+    cur = torch.__version__.split("+")[0]
+
+    class Ext:
+        def __init__(self, t=cur, bh=22707, rt=22707):
+            self._t, self._bh, self._rt = t, bh, rt
+
+        def build_info(self):
+            return {"torch": self._t, "rccl_header": self._bh,
+                    "glibcxx_use_cxx11_abi": 1}
+
+        def version_info(self):
+            return {"rccl": self._rt, "hip_runtime": 0}
+
+    monkeypatch.delenv("MPI4JAX_AMD_SKIP_ABI_CHECK", raising=False)
+    _check_build_compat(Ext())  # exact match: fine
+    with pytest.raises(ImportError, match="C\\+\\+ ABI"):
+        _check_build_compat(Ext(t="9.9.0"))
+    with pytest.raises(ImportError, match="major-version"):
+        _check_build_compat(Ext(bh=32707))
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        _check_build_compat(Ext(rt=22606))  # the torch-lib librccl skew
+        assert any("skew" in str(x.message) for x in w)
+    monkeypatch.setenv("MPI4JAX_AMD_SKIP_ABI_CHECK", "1")
+    _check_build_compat(Ext(t="9.9.0"))  # bypassed
