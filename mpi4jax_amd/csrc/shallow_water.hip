@@ -1265,20 +1265,13 @@ __global__ void sw_stage30_ringA(SwArgs<float> a) {
   a.fn[idx] = r8.vv;
 }
 
-__global__ void sw_stage30_ringB(SwArgs<float> a) {
+// the stage7_cell friction (division math) for one interior ring cell,
+// inputs from the staged u'/v' strip — used for a cell's own final AND
+// recomputed (bitwise-identically) when that cell is the wrap source of
+// a halo cell, so ringB can write the end-of-step halo refresh itself
+__device__ inline void ringb_final_uv(const SwArgs<float>& a, int j,
+                                      int i, float* uo, float* vo) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  int j, i;
-  if (!s30_ring_cell(a, (long long)blockIdx.x * blockDim.x + threadIdx.x,
-                     &j, &i))
-    return;
-  const long long idx = (long long)j * nx + i;
-  if (s30_fast_out(a, j, i & ~1)) return;  // fast kernel owns it
-  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
-    a.u2[idx] = a.u[idx];  // pass-through; open halos are overwritten
-    a.v2[idx] = a.v[idx];  // by the end-of-step wrap refresh
-    return;
-  }
-  // stage7_cell math on the staged u'/v' strip
   float uc, vc, un, vn, us, vs, ue, ve, uw, vw;
   s30_read_uv(a, j, i, &uc, &vc);
   s30_read_uv(a, j + 1, i, &un, &vn);
@@ -1303,6 +1296,48 @@ __global__ void sw_stage30_ringB(SwArgs<float> a) {
   float vv2 = vc + a.dt * lv;
   if (a.f.east_wall && i == nx - 2) uu2 = 0.f;
   if (a.f.north_wall && j == ny - 2) vv2 = 0.f;
+  *uo = uu2;
+  *vo = vv2;
+}
+
+__global__ void sw_stage30_ringB(SwArgs<float> a) {
+  const int ny = (int)a.ny, nx = (int)a.nx;
+  int j, i;
+  if (!s30_ring_cell(a, (long long)blockIdx.x * blockDim.x + threadIdx.x,
+                     &j, &i))
+    return;
+  const long long idx = (long long)j * nx + i;
+  if (s30_fast_out(a, j, i & ~1)) return;  // fast kernel owns it
+  if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
+    // halo cell: write what the end-of-step exchange would leave, so
+    // the fused step needs NO separate wrap/pack launches.  Open axes
+    // map to their periodic wrap source (both at once = the exchange's
+    // "corners win" order); a source that is itself a halo cell means
+    // its value is the pass-through of the pre-update field there.
+    int js = j, is = i;
+    if (j == 0 && a.f.south_open) js = ny - 2;
+    else if (j == ny - 1 && a.f.north_open) js = 1;
+    if (i == 0 && a.f.west_open) is = nx - 2;
+    else if (i == nx - 1 && a.f.east_open) is = 1;
+    const long long sdx = (long long)js * nx + is;
+    if (js < 1 || js > ny - 2 || is < 1 || is > nx - 2) {
+      a.h2[idx] = a.h[sdx];  // closed-axis source: pass-through
+      a.u2[idx] = a.u[sdx];
+      a.v2[idx] = a.v[sdx];
+    } else {
+      a.h2[idx] = a.h2[sdx];  // update output, stored by a prior kernel
+      // the wrap source is never s30_fast_out (cols 1/nx-2, rows
+      // 1/ny-2 all fail its bounds), so recompute its ringB final —
+      // bitwise-identical to what its own thread stores
+      float uu2, vv2;
+      ringb_final_uv(a, js, is, &uu2, &vv2);
+      a.u2[idx] = uu2;
+      a.v2[idx] = vv2;
+    }
+    return;
+  }
+  float uu2, vv2;
+  ringb_final_uv(a, j, i, &uu2, &vv2);
   a.u2[idx] = uu2;
   a.v2[idx] = vv2;
 }

@@ -315,12 +315,12 @@ class ShallowWater:
         if s1 is not None:
             stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
         if s6 == 30:
-            # fused update+friction: u'/v' never round-trip through HBM
-            # and the mid-step exchange disappears; one wrap refresh of
-            # the finals closes the step
+            # fused update+friction: u'/v' never round-trip through HBM,
+            # the mid-step exchange disappears, and the ring kernel
+            # writes the end-of-step wrap refresh itself — a complete
+            # model step with zero exchange launches
             stage(s6)
             self._swap("h", "u", "v")
-            self._exchange_fields([fb["h"], fb["u"], fb["v"]])
             for k in ("h", "u", "v"):
                 fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
             return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"],
