@@ -1233,13 +1233,15 @@ __device__ inline bool s30_ring_cell(const SwArgs<float>& a, long long t,
   if (t >= top + (long long)mid * edge_cols) return false;
   if (t < top) {
     const int r = (int)(t / nx);
-    *jo = r < 4 ? r : ny - 8 + r;
+    // ny < 8: edge_rows == ny and r IS the row (the two-band formula
+    // would alias low rows and skip the top ones)
+    *jo = (r < 4 || ny < 8) ? r : ny - 8 + r;
     *io = (int)(t % nx);
   } else {
     const long long u = t - top;
     *jo = 4 + (int)(u / edge_cols);
     const int c = (int)(u % edge_cols);
-    *io = c < 6 ? c : nx - 12 + c;
+    *io = (c < 6 || nx < 12) ? c : nx - 12 + c;
   }
   return true;
 }

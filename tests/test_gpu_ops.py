@@ -227,7 +227,7 @@ def test_shallow_water_gpu_step():
 
 
 @pytest.mark.parametrize("nx,ny", [(120, 60), (53, 37), (511, 130),
-                                   (1040, 24)])
+                                   (1040, 24), (64, 5), (8, 32)])
 def test_stage30_matches_two_kernel_path(nx, ny, monkeypatch):
     """The fused update+friction kernel pair (stage 30 fast + cleanup)
     must reproduce the two-kernel pipeline (stage 19 + wrap exchange +
