@@ -1151,7 +1151,8 @@ __device__ inline bool s30_fast_out(const SwArgs<float>& a, int j,
          i0 + 4 <= imax && i0 + 5 < nx;
 }
 
-__global__ void sw_stage30v(SwArgs<float> a) {
+template <int NT_>
+__global__ void __launch_bounds__(NT_) sw_stage30t(SwArgs<float> a) {
   const int ny = (int)a.ny, nx = (int)a.nx;
   const int bpc = (nx + 2 * kPC30 - 1) / (2 * kPC30);  // blocks per band
   const int T_ = (int)gridDim.x;
@@ -1165,7 +1166,7 @@ __global__ void sw_stage30v(SwArgs<float> a) {
   const int tid = (int)threadIdx.x;
 
   __shared__ float su[kTR30][2 * kTP30], sv[kTR30][2 * kTP30];
-  for (int t = tid; t < kFill30; t += 256) {
+  for (int t = tid; t < kFill30; t += NT_) {
     const int lr = t / kTP30, lp = t % kTP30;
     const int rr = jb - 1 + lr;
     const int i0 = cb - 2 + 2 * lp;
@@ -1189,7 +1190,7 @@ __global__ void sw_stage30v(SwArgs<float> a) {
   }
   __syncthreads();
 
-  for (int t = tid; t < kTJ30 * kPC30; t += 256) {
+  for (int t = tid; t < kTJ30 * kPC30; t += NT_) {
     const int ro = t / kPC30, po = t % kPC30;
     const int j = jb + ro;
     const int i0 = cb + 2 * po;
@@ -1432,11 +1433,17 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 22: hipLaunchKernelGGL(sw_stage21t<8>, grid, block, 0, stream, a); break;
       case 23: hipLaunchKernelGGL(sw_stage21t<16>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
-      case 30: {
+      case 30:
+      case 31: {  // 31 = 512-thread blocks (single-round LDS fill)
         long long bpc = (p.nx + 2 * kPC30 - 1) / (2 * kPC30);
         long long bands = (p.ny + kTJ30 - 1) / kTJ30;
         dim3 g30((unsigned)(bpc * bands));
-        hipLaunchKernelGGL(sw_stage30v, g30, block, 0, stream, a);
+        if (stage == 31) {
+          hipLaunchKernelGGL(sw_stage30t<512>, g30, dim3(512), 0, stream,
+                             a);
+        } else {
+          hipLaunchKernelGGL(sw_stage30t<256>, g30, block, 0, stream, a);
+        }
         long long er = p.ny < 8 ? p.ny : 8, ec = p.nx < 12 ? p.nx : 12;
         long long ring = er * p.nx +
                          (p.ny > 8 ? (p.ny - 8) * ec : 0);

@@ -276,6 +276,8 @@ class ShallowWater:
                 # path to FMA-contraction rounding (tests/test_gpu_ops.py
                 # ::test_stage30_matches_two_kernel_path).  s7=None marks
                 # the fused-friction step shape.
+                if os.environ.get("MPI4JAX_AMD_SW_FUSE512") == "1":
+                    return None, 31, None  # 512-thread single-round fill
                 return None, 30, None
             # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
             # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
@@ -314,7 +316,7 @@ class ShallowWater:
 
         if s1 is not None:
             stage(s1)     # fe, fn, q, ke (with open-edge halo formulas)
-        if s6 == 30:
+        if s6 in (30, 31):
             # fused update+friction: u'/v' never round-trip through HBM,
             # the mid-step exchange disappears, and the ring kernel
             # writes the end-of-step wrap refresh itself — a complete
