@@ -629,7 +629,7 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
               double cor_dj, double ab_a, double ab_b,
               std::vector<int64_t> flags) {
   TORCH_CHECK(bufs.size() == 16, "sw_stage expects 16 buffers");
-  TORCH_CHECK(flags.size() == 6, "sw_stage expects 6 flags");
+  TORCH_CHECK(flags.size() == 7, "sw_stage expects 7 flags");
   const at::Tensor& h = bufs[4];
   TORCH_CHECK(h.is_cuda() && h.is_contiguous() && h.dim() == 2,
               "bad shallow-water state tensor");
@@ -663,6 +663,7 @@ void sw_stage(int64_t stage, std::vector<at::Tensor> bufs, double dx,
   p.east_open = (int)flags[3];
   p.east_wall = (int)flags[4];
   p.north_wall = (int)flags[5];
+  p.x_wrap = (int)flags[6];
   launch_sw_stage((int)stage, p, is_double ? 1 : 0, cur_stream());
 }
 

@@ -55,6 +55,7 @@ struct SwLaunchParams {
   double ab_a, ab_b;
   int south_open, north_open, west_open, east_open;
   int east_wall, north_wall;
+  int x_wrap;  // x halos are LOCAL periodic wraps (nproc_x==1), not remote
 };
 
 void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,

@@ -48,6 +48,9 @@ struct SwFlags {
   // physical walls (reference shallow_water.py:258-262): u column nx-2
   // zeroed on a closed east edge, v row ny-2 zeroed on the north edge
   int east_wall, north_wall;
+  // x halos are LOCAL periodic wraps (nproc_x == 1 && periodic), not
+  // remote exchanges — lets the fused ring kernel synthesize them
+  int x_wrap;
 };
 
 template <typename T>
@@ -1101,24 +1104,29 @@ __device__ inline bool s30_in_ring(const SwArgs<float>& a, int jj,
 }
 
 // what the ring friction reads at (jj, ii): the mid-step exchange's
-// value — open-edge halos wrap to their (world-1) periodic source,
-// closed-edge halos pass the pre-update field through, interior cells
-// read the stored u'/v' strip.  World>1 never runs stage 30 (remote
-// halo values cannot be synthesized locally).
+// value.  Local periodic x halos (x_wrap) map to their wrap source;
+// remote open halos read the EXCHANGED fe/fn strip (the N>1 fused step
+// runs a real fe/fn halo exchange between ringA and ringB);
+// closed-edge halos pass the pre-update field through; interior cells
+// read the staged strip.
 __device__ inline void s30_read_uv(const SwArgs<float>& a, int jj, int ii,
                                    float* uu, float* vv) {
   const int ny = (int)a.ny, nx = (int)a.nx;
-  int js = jj, is = ii;
-  if (js == 0 && a.f.south_open) js = ny - 2;
-  else if (js == ny - 1 && a.f.north_open) js = 1;
-  if (is == 0 && a.f.west_open) is = nx - 2;
-  else if (is == nx - 1 && a.f.east_open) is = 1;
-  const long long idx = (long long)js * nx + is;
-  if (js < 1 || js > ny - 2 || is < 1 || is > nx - 2) {
+  int is = ii;
+  if (a.f.x_wrap) {
+    if (ii == 0) is = nx - 2;
+    else if (ii == nx - 1) is = 1;
+  }
+  const long long idx = (long long)jj * nx + is;
+  const bool closed = (jj == 0 && !a.f.south_open) ||
+                      (jj == ny - 1 && !a.f.north_open) ||
+                      (is == 0 && !a.f.west_open) ||
+                      (is == nx - 1 && !a.f.east_open);
+  if (closed) {
     *uu = a.u[idx];  // closed-axis halo: pass-through
     *vv = a.v[idx];
   } else {
-    *uu = a.fe[idx];  // u'/v' strip staged by stage30v / ringA
+    *uu = a.fe[idx];  // staged (interior) or exchanged (remote halo)
     *vv = a.fn[idx];
   }
 }
@@ -1312,28 +1320,29 @@ __global__ void sw_stage30_ringB(SwArgs<float> a) {
   const long long idx = (long long)j * nx + i;
   if (s30_fast_out(a, j, i & ~1)) return;  // fast kernel owns it
   if (j < 1 || j > ny - 2 || i < 1 || i > nx - 2) {
-    // halo cell: write what the end-of-step exchange would leave, so
-    // the fused step needs NO separate wrap/pack launches.  Open axes
-    // map to their periodic wrap source (both at once = the exchange's
-    // "corners win" order); a source that is itself a halo cell means
-    // its value is the pass-through of the pre-update field there.
-    int js = j, is = i;
-    if (j == 0 && a.f.south_open) js = ny - 2;
-    else if (j == ny - 1 && a.f.north_open) js = 1;
-    if (i == 0 && a.f.west_open) is = nx - 2;
-    else if (i == nx - 1 && a.f.east_open) is = 1;
-    const long long sdx = (long long)js * nx + is;
-    if (js < 1 || js > ny - 2 || is < 1 || is > nx - 2) {
-      a.h2[idx] = a.h[sdx];  // closed-axis source: pass-through
+    // halo cell.  LOCAL periodic x halos (x_wrap): write what the
+    // end-of-step wrap would leave, so the world-1 fused step needs NO
+    // separate wrap/pack launches.  Remote-open halos: pass-through —
+    // the real end-of-step exchange overwrites them.  Closed halos:
+    // pass-through is the final value.  A wrap source that is itself a
+    // (closed-y) halo cell passes the pre-update field through.
+    int is = i;
+    if (a.f.x_wrap) {
+      if (i == 0) is = nx - 2;
+      else if (i == nx - 1) is = 1;
+    }
+    const long long sdx = (long long)j * nx + is;
+    if (is == i || j < 1 || j > ny - 2) {
+      a.h2[idx] = a.h[sdx];  // pass-through (possibly x-wrapped corner)
       a.u2[idx] = a.u[sdx];
       a.v2[idx] = a.v[sdx];
     } else {
       a.h2[idx] = a.h2[sdx];  // update output, stored by a prior kernel
-      // the wrap source is never s30_fast_out (cols 1/nx-2, rows
-      // 1/ny-2 all fail its bounds), so recompute its ringB final —
-      // bitwise-identical to what its own thread stores
+      // the wrap source is never s30_fast_out (cols 1/nx-2 fail its
+      // bounds), so recompute its ringB final — bitwise-identical to
+      // what its own thread stores
       float uu2, vv2;
-      ringb_final_uv(a, js, is, &uu2, &vv2);
+      ringb_final_uv(a, j, is, &uu2, &vv2);
       a.u2[idx] = uu2;
       a.v2[idx] = vv2;
     }
@@ -1386,7 +1395,7 @@ static void sw_fill_args(SwArgs<T>& a, const SwLaunchParams& p) {
   a.ab_a = (T)p.ab_a;
   a.ab_b = (T)p.ab_b;
   a.f = {p.south_open, p.north_open, p.west_open, p.east_open, p.east_wall,
-         p.north_wall};
+         p.north_wall, p.x_wrap};
 }
 
 template <typename T>
@@ -1434,7 +1443,9 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
       case 23: hipLaunchKernelGGL(sw_stage21t<16>, grid, block, 0, stream, a); break;
       case 27: hipLaunchKernelGGL(sw_stage27v, grid, block, 0, stream, a); break;
       case 30:
-      case 31: {  // 31 = 512-thread blocks (single-round LDS fill)
+      case 31:    // 31 = 512-thread blocks (single-round LDS fill)
+      case 32: {  // 32 = fast + ringA only (N>1: a real fe/fn halo
+                  // exchange runs before ringB, launched as stage 33)
         long long bpc = (p.nx + 2 * kPC30 - 1) / (2 * kPC30);
         long long bands = (p.ny + kTJ30 - 1) / kTJ30;
         dim3 g30((unsigned)(bpc * bands));
@@ -1449,6 +1460,16 @@ void launch_sw_stage(int stage, const SwLaunchParams& p, int is_double,
                          (p.ny > 8 ? (p.ny - 8) * ec : 0);
         dim3 gcl((unsigned)((ring + 255) / 256));
         hipLaunchKernelGGL(sw_stage30_ringA, gcl, block, 0, stream, a);
+        if (stage != 32) {
+          hipLaunchKernelGGL(sw_stage30_ringB, gcl, block, 0, stream, a);
+        }
+        break;
+      }
+      case 33: {  // ringB alone, after the fe/fn halo exchange
+        long long er = p.ny < 8 ? p.ny : 8, ec = p.nx < 12 ? p.nx : 12;
+        long long ring = er * p.nx +
+                         (p.ny > 8 ? (p.ny - 8) * ec : 0);
+        dim3 gcl((unsigned)((ring + 255) / 256));
         hipLaunchKernelGGL(sw_stage30_ringB, gcl, block, 0, stream, a);
         break;
       }

@@ -209,6 +209,9 @@ class ShallowWater:
             int(g.neighbor("east") is not None),
             int((not self.periodic_x) and self._at_east_edge()),
             int(self._at_north_edge()),
+            # x halos are LOCAL periodic wraps, not remote exchanges
+            int(self.periodic_x and g.nproc_x == 1
+                and not self._force_remote_exchange),
         ]
 
     def _init_fused_buffers(self, state):
@@ -265,20 +268,22 @@ class ShallowWater:
                 # band-tiled tendency kernel: TJ rows per 256-thread block
                 return None, {"4": 21, "8": 22, "16": 23}[tile], 27
             if (self.lateral_viscosity > 0
-                    and self.grid.nproc_y * self.grid.nproc_x == 1
-                    and not self._force_remote_exchange
                     and not os.environ.get("MPI4JAX_AMD_SW_OVERLAP")
                     and os.environ.get("MPI4JAX_AMD_SW_NOFUSE") != "1"):
-                # single-rank fully-local halos: update+friction fused
-                # (stage 30 fast kernel + boundary cleanup) — drops the
-                # u'/v' intermediate round trip (4 of 16 field passes)
-                # and the mid-step exchange; equivalent to the two-kernel
-                # path to FMA-contraction rounding (tests/test_gpu_ops.py
-                # ::test_stage30_matches_two_kernel_path).  s7=None marks
-                # the fused-friction step shape.
-                if os.environ.get("MPI4JAX_AMD_SW_FUSE512") == "1":
-                    return None, 31, None  # 512-thread single-round fill
-                return None, 30, None
+                # update+friction fused (stage 30 fast kernel + boundary
+                # ring kernels) — drops the u'/v' intermediate round
+                # trip (4 of 16 field passes); equivalent to the
+                # two-kernel path to FMA-contraction rounding
+                # (tests/test_gpu_ops.py::test_stage30_*).  s7=None
+                # marks the world-1 shape (in-kernel wrap refresh, zero
+                # exchange launches); s7=33 the multi-rank shape (a real
+                # fe/fn strip exchange between ringA and ringB).
+                if (self.grid.nproc_y * self.grid.nproc_x == 1
+                        and not self._force_remote_exchange):
+                    if os.environ.get("MPI4JAX_AMD_SW_FUSE512") == "1":
+                        return None, 31, None  # 512-thread variant
+                    return None, 30, None
+                return None, 32, 33
             # 2-col merged single pass: 68 VGPRs -> 7 waves/SIMD, measured
             # fastest (stage18v parks ~48% of cycles on memory at 4 waves)
             return None, 19, 27
@@ -323,6 +328,20 @@ class ShallowWater:
             # model step with zero exchange launches
             stage(s6)
             self._swap("h", "u", "v")
+            for k in ("h", "u", "v"):
+                fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
+            return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"],
+                              fb["do_u"], fb["do_v"])
+        if s6 == 32:
+            # multi-rank fused step: the friction's only remote need is
+            # the u'/v' boundary strip (staged in fe/fn) — exchange
+            # those two fields between the ring kernels, then refresh
+            # the finals' halos once at the end
+            stage(s6)     # fast kernel + ringA (completes the strip)
+            self._exchange_fields([fb["fe"], fb["fn"]])
+            stage(s7)     # ringB: ring friction from the strip
+            self._swap("h", "u", "v")
+            self._exchange_fields([fb["h"], fb["u"], fb["v"]])
             for k in ("h", "u", "v"):
                 fb[f"do_{k}"], fb[f"dn_{k}"] = fb[f"dn_{k}"], fb[f"do_{k}"]
             return ModelState(fb["h"], fb["u"], fb["v"], fb["do_h"],

@@ -718,3 +718,63 @@ def test_gpu_tag_hard_failure(monkeypatch):
     got42 = m.recv(x, 0, tag=42)
     torch.cuda.synchronize()
     assert got43[0].item() == 2.0 and got42[0].item() == 1.0
+
+
+@pytest.mark.parametrize("nx,ny", [(120, 60), (53, 37)])
+def test_stage32_multirank_fused_path_forced_remote(nx, ny, monkeypatch):
+    """The multi-rank fused step (stage 32 fast+ringA, real fe/fn strip
+    exchange, stage 33 ringB) vs the two-kernel pipeline — both with the
+    world-1 periodic wraps expressed as REAL RCCL self-transfers
+    (`_force_remote_exchange`), so the exact code path the driver's
+    multi-GPU run takes (pack → grouped p2p → unpack between the ring
+    kernels) executes against real transport matching."""
+    from mpi4jax_amd.models import ShallowWater
+
+    results = {}
+    monkeypatch.delenv("MPI4JAX_AMD_SW_FUSE512", raising=False)
+    for nofuse in ("1", ""):
+        if nofuse:
+            monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", nofuse)
+        else:
+            monkeypatch.delenv("MPI4JAX_AMD_SW_NOFUSE", raising=False)
+        sw = ShallowWater(nx=nx, ny=ny, device="cuda", fused=True,
+                          _force_remote_exchange=True,
+                          comm=m.get_world().Clone())
+        assert sw._stage_plan()[1] == (19 if nofuse else 32)
+        st = sw.step(sw.initial_conditions(), first_step=True)
+        for _ in range(8):
+            st = sw.step(st)
+        torch.cuda.synchronize()
+        results[nofuse] = {k: getattr(st, k).clone()
+                           for k in ("h", "u", "v")}
+    for k in ("h", "u", "v"):
+        a, b = results["1"][k], results[""][k]
+        assert torch.allclose(a, b, atol=5e-5, rtol=1e-4), (
+            k, (a - b).abs().max().item())
+
+
+def test_stage32_matches_stage30_world1(monkeypatch):
+    """At world 1 the multi-rank fused shape (32/33, forced-remote
+    strip exchange) and the zero-exchange shape (30) must agree — the
+    strip values travel by RCCL instead of being synthesized, nothing
+    else differs."""
+    from mpi4jax_amd.models import ShallowWater
+
+    monkeypatch.delenv("MPI4JAX_AMD_SW_NOFUSE", raising=False)
+    monkeypatch.delenv("MPI4JAX_AMD_SW_FUSE512", raising=False)
+    results = {}
+    for forced in (False, True):
+        sw = ShallowWater(nx=130, ny=66, device="cuda", fused=True,
+                          _force_remote_exchange=forced,
+                          comm=m.get_world().Clone())
+        assert sw._stage_plan()[1] == (32 if forced else 30)
+        st = sw.step(sw.initial_conditions(), first_step=True)
+        for _ in range(6):
+            st = sw.step(st)
+        torch.cuda.synchronize()
+        results[forced] = {k: getattr(st, k).clone()
+                           for k in ("h", "u", "v")}
+    for k in ("h", "u", "v"):
+        a, b = results[False][k], results[True][k]
+        assert torch.allclose(a, b, atol=2e-6, rtol=1e-6), (
+            k, (a - b).abs().max().item())

@@ -206,10 +206,11 @@ def test_stage_plan_selection(monkeypatch):
     monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", "1")
     assert sw._stage_plan() == (None, 19, 27)
     monkeypatch.delenv("MPI4JAX_AMD_SW_NOFUSE")
-    # remote halos (forced) disqualify stage 30
+    # remote halos (forced) select the multi-rank fused shape: fast
+    # kernel + ringA (32), real fe/fn strip exchange, ringB (33)
     sw_r = ShallowWater(nx=12, ny=6, device="cpu",
                         _force_remote_exchange=True)
-    assert sw_r._stage_plan() == (None, 19, 27)
+    assert sw_r._stage_plan() == (None, 32, 33)
     monkeypatch.setenv("MPI4JAX_AMD_SW_NOFUSE", "1")
     monkeypatch.setenv("MPI4JAX_AMD_SW_4COL", "1")
     assert sw._stage_plan() == (None, 18, 17)
