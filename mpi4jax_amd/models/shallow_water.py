@@ -13,11 +13,16 @@ The reference's benchmark config (docs/shallow-water.rst:49-52) is a
 day is the headline metric.
 
 Two step implementations share semantics: an eager torch-op path (CPU
-and reference/debugging) and the fused MI355X path — two or three CDNA4
-kernels (stage 19: tendencies+update; stage 27: friction; selectable
-variants and scalar fallbacks) plus a one-native-call halo exchange
-(merged pack launch, single RCCL group, merged unpack launch), with
-validated hipGraph capture for multistep replay at any world size.
+and reference/debugging) and the fused MI355X path — by default ONE
+LDS-tiled update+friction pass plus two boundary-ring kernels (stage
+30/32; at world 1 the ring kernel also writes the periodic wrap refresh,
+so a whole model step launches zero exchange kernels; multi-rank, a
+two-field u'/v'-strip exchange runs between the ring kernels), with the
+two-kernel pipeline (stage 19 tendencies+update, stage 27 friction,
+selectable variants and scalar fallbacks) behind MPI4JAX_AMD_SW_NOFUSE,
+a one-native-call halo exchange (merged pack launch, single RCCL group,
+merged unpack launch), and validated hipGraph capture for multistep
+replay at any world size.
 """
 
 import math
