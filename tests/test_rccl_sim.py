@@ -113,3 +113,62 @@ def _strict_worker(rank, ws):
 
 def test_gpu_tags_fail_loudly_without_envelope():
     run_multiproc(_strict_worker, 2)
+
+
+def _random_worker(rank, ws, seed):
+    """Property test: random tags/sizes/recv strategies at world 2 must
+    follow MPI matching — a recv by (source, tag) delivers the
+    EARLIEST-sent unconsumed matching message; ANY_TAG/ANY_SOURCE
+    deliver the earliest unconsumed message outright (the stash is
+    always older than the queue, so claim()'s stash-first order is the
+    send order)."""
+    import random
+
+    import mpi4jax_amd as m
+    from mpi4jax_amd._backend import rccl
+    from mpi4jax_amd.utils.status import ANY_SOURCE, ANY_TAG, Status
+
+    os.environ["MPI4JAX_AMD_GPU_ENVELOPE"] = "1"
+    comm = m.get_world()
+    fake = _FakeRcclExt(comm)
+    rccl._EXT = fake
+    rccl._handle = lambda c: 0
+
+    rng = random.Random(seed)
+    n_msgs = 14
+    msgs = [(k, rng.randrange(0, 4), rng.randrange(1, 9))
+            for k in range(n_msgs)]  # (serial, tag, numel)
+    if rank == 0:
+        for serial, tag, numel in msgs:
+            rccl.send(torch.full((numel,), float(serial)), 1, tag, comm)
+        fake.flush()
+    else:
+        remaining = list(msgs)
+        while remaining:
+            mode = rng.randrange(3)
+            if mode == 0:          # specific tag (of a remaining msg)
+                tag = rng.choice(remaining)[1]
+                src = 0
+            elif mode == 1:        # ANY_TAG
+                tag, src = ANY_TAG, 0
+            else:                  # full wildcard
+                tag, src = ANY_TAG, ANY_SOURCE
+            exp = next(mm for mm in remaining
+                       if tag in (ANY_TAG, mm[1]))
+            remaining.remove(exp)
+            st = Status()
+            out = rccl.recv(torch.empty(exp[2]), src, tag, comm, st)
+            assert torch.equal(out, torch.full((exp[2],),
+                                               float(exp[0]))), (
+                exp, out[0].item())
+            assert (st.source, st.tag) == (0, exp[1])
+        from mpi4jax_amd._backend.envelope import box_for
+
+        box = box_for(comm)
+        assert box.queue == [] and box.stash == []
+    dist.barrier(group=comm.gloo_group)
+
+
+@pytest.mark.parametrize("seed", [7, 23, 1009])
+def test_rccl_envelope_random_matching_world2(seed):
+    run_multiproc(_random_worker, 2, args=(seed,))
