@@ -1083,8 +1083,8 @@ __global__ void sw_stage21t(SwArgs<float> a) {
 // the friction to the TJ middle rows from LDS.
 //
 // Numerics: same formulas in the same order as the two-kernel path
-// (stage19_math / stage8_math / stage27's rdx form), with the wrap
-// exchange synthesized by uvprime_cell.  NOT bitwise: the compiler
+// (stage19_math / stage8_math / stage27's rdx form), with local wrap
+// halos synthesized via s30_read_uv.  NOT bitwise: the compiler
 // contracts the shared expression trees to FMA differently per inlining
 // site (~1 ulp/step, tolerance-tested in tests/test_gpu_ops.py::
 // test_stage30_matches_two_kernel_path).
