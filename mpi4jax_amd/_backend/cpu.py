@@ -12,6 +12,7 @@ import weakref
 import torch
 import torch.distributed as dist
 
+from . import envelope
 from ..ops.reduce_ops import Op, GLOO_OP_MAP, combine
 from ..utils.status import ANY_SOURCE, ANY_TAG
 from ..utils.logging import debug_timer
@@ -20,6 +21,53 @@ from ..utils.logging import debug_timer
 # MPI supports buffered self sends; gloo does not, so we emulate locally.
 # Weak-keyed so a GC'd communicator can never alias a new one's queue.
 _SELF_QUEUES = weakref.WeakKeyDictionary()
+
+# Remote p2p rides the SAME envelope plane as the RCCL backend
+# (``_backend/envelope.py``): data travels on one fixed wire tag (FIFO
+# per ordered pair — exactly RCCL's matching), envelopes carry
+# ``(tag, nbytes)``, and ``claim()`` provides full MPI matching —
+# earliest-match by (source, tag), ANY_SOURCE/ANY_TAG wildcards, and
+# Status synthesis from the envelope (gloo's own tag matching cannot
+# express ANY_TAG, and it reorders delivery relative to MPI's
+# earliest-send rule).  Data sends are non-blocking so a differently-
+# tagged message ahead in the pipe can never head-block the sender.
+WIRE_TAG = envelope.ENV_TAG + 1
+
+_PENDING = weakref.WeakKeyDictionary()  # comm -> [(isend work, buffer)]
+
+
+def _isend_data(comm, wire, dest):
+    w = dist.isend(wire, dst=comm.global_rank(dest),
+                   group=comm.gloo_group, tag=WIRE_TAG)
+    pl = _PENDING.setdefault(comm, [])
+    pl.append((w, wire))
+    pl[:] = [(ww, b) for ww, b in pl if not ww.is_completed()]
+
+
+def drain_pending():
+    """Wait for all in-flight data sends (MPI_Finalize semantics: blocks
+    until every send has been matched and delivered)."""
+    for pl in _PENDING.values():
+        for w, _ in pl:
+            w.wait()
+        pl.clear()
+
+
+def _env_recv_into(out, source, tag, comm, status):
+    def recv_bytes(src, nbytes):
+        buf = torch.empty(nbytes, dtype=torch.uint8)
+        dist.recv(buf, src=comm.global_rank(src), group=comm.gloo_group,
+                  tag=WIRE_TAG)
+        return buf
+
+    s, t, data = envelope.box_for(comm).claim(source, tag, recv_bytes)
+    if data is not None:  # drained ahead of order into the stash
+        out.copy_(data.view(out.dtype).reshape(out.shape))
+    else:
+        dist.recv(_wire(out), src=comm.global_rank(s),
+                  group=comm.gloo_group, tag=WIRE_TAG)
+    _fill_status(status, s, t, out)
+    return out
 
 
 def _self_queue(comm):
@@ -206,14 +254,21 @@ def scan(x, op, comm):
         return out
 
 
+def _check_send_tag(tag, what):
+    if tag == ANY_TAG:
+        raise ValueError(f"{what}: ANY_TAG is not a sendable tag")
+
+
 def send(x, dest, tag, comm):
     with debug_timer("Send", comm.rank, f"to {dest}, tag {tag}"):
+        _check_send_tag(tag, "send")
         xc = x.contiguous().clone()
         if dest == comm.rank:
             _self_queue(comm).append((tag, xc))
             return
-        dist.send(_wire(xc), dst=comm.global_rank(dest),
-                  group=comm.gloo_group, tag=max(tag, 0))
+        envelope.box_for(comm).post(dest, tag,
+                                    xc.numel() * xc.element_size())
+        _isend_data(comm, _wire(xc), dest)
 
 
 def recv(template, source, tag, comm, status):
@@ -232,15 +287,7 @@ def recv(template, source, tag, comm, status):
             raise RuntimeError(
                 "recv from self with no matching buffered send"
             )
-        src = None if source == ANY_SOURCE else comm.global_rank(source)
-        sender = dist.recv(_wire(out), src=src, group=comm.gloo_group,
-                           tag=max(tag, 0))
-        src_comm_rank = (
-            source if source != ANY_SOURCE
-            else comm._ranks.index(sender)
-        )
-        _fill_status(status, src_comm_rank, tag, out)
-        return out
+        return _env_recv_into(out, source, tag, comm, status)
 
 
 def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
@@ -252,21 +299,30 @@ def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
             out.copy_(sendbuf.reshape(out.shape))
             _fill_status(status, source, recvtag, out)
             return out
-        sc = sendbuf.contiguous()
-        sreq = dist.P2POp(
-            dist.isend, _wire(sc), peer=comm.global_rank(dest),
-            group=comm.gloo_group, tag=max(sendtag, 0),
-        )
-        rreq = dist.P2POp(
-            dist.irecv, _wire(out), peer=comm.global_rank(source),
-            group=comm.gloo_group, tag=max(recvtag, 0),
-        )
-        # batched non-blocking pair: deadlock-free regardless of order
-        reqs = dist.batch_isend_irecv([sreq, rreq])
-        for r in reqs:
-            r.wait()
-        _fill_status(status, source, recvtag, out)
-        return out
+        sc = sendbuf.contiguous().clone()
+        # same envelope plane as send/recv, so a sendrecv pairs with a
+        # plain send or recv on the peer (the halo oracle does exactly
+        # that); the data isend never blocks, so any issue order is
+        # deadlock-free
+        if dest == comm.rank:
+            _self_queue(comm).append((sendtag, sc))
+        else:
+            _check_send_tag(sendtag, "sendrecv(sendtag)")
+            envelope.box_for(comm).post(dest, sendtag,
+                                        sc.numel() * sc.element_size())
+            _isend_data(comm, _wire(sc), dest)
+        if source == comm.rank:
+            q = _self_queue(comm)
+            for i, (t, buf) in enumerate(q):
+                if recvtag in (ANY_TAG, t):
+                    q.pop(i)
+                    out.copy_(buf.reshape(out.shape))
+                    _fill_status(status, comm.rank, t, out)
+                    return out
+            raise RuntimeError(
+                "sendrecv from self with no matching buffered send"
+            )
+        return _env_recv_into(out, source, recvtag, comm, status)
 
 
 def _fill_status(status, source, tag, out):

@@ -357,6 +357,8 @@ def send(x, dest, tag, comm):
         from . import envelope
 
         if envelope.enabled() and comm.gloo_group is not None:
+            if tag == ANY_TAG:
+                raise ValueError("send: ANY_TAG is not a sendable tag")
             envelope.box_for(comm).post(dest, tag,
                                         xc.numel() * xc.element_size())
         else:

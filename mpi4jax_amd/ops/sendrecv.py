@@ -35,11 +35,14 @@ class _Sendrecv(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad):
         # VJP: route the output cotangent back along the reversed edge
-        # (sendrecv.py:278-293 swaps source and dest).
-        source, dest, sendtag, recvtag, comm, backend = ctx.meta
+        # (sendrecv.py:278-293 swaps source and dest).  Tags do not
+        # identify cotangents — and the forward recvtag may be ANY_TAG,
+        # which is not a sendable tag — so the backward pair always uses
+        # (sendtag=0, recvtag=ANY_TAG): both sides agree by construction.
+        source, dest, _sendtag, _recvtag, comm, backend = ctx.meta
         template = grad.new_empty(ctx.send_shape)
         grad_send = backend.sendrecv(
-            grad.contiguous(), template, dest, source, recvtag, sendtag,
+            grad.contiguous(), template, dest, source, 0, ANY_TAG,
             comm, None,
         )
         return grad_send, None, None, None, None, None, None, None, None

@@ -285,6 +285,9 @@ def flush():
     try:
         if torch.cuda.is_available() and torch.cuda.is_initialized():
             torch.cuda.synchronize()
+        from .._backend import cpu as _cpu
+
+        _cpu.drain_pending()  # MPI_Finalize semantics for in-flight sends
         from .._backend import rccl
 
         if rccl.ext_is_loaded():

@@ -49,3 +49,15 @@ def test_tensor_parallel_example_two_ranks():
     )
     assert res.returncode == 0, res.stderr + res.stdout
     assert "OK: 6 TP steps over 2 rank(s)" in res.stdout
+
+
+def test_master_worker_example_three_ranks():
+    """Dynamic task farm: ANY_SOURCE + ANY_TAG + Status-driven work
+    hand-out (the envelope plane's flagship CPU use)."""
+    res = subprocess.run(
+        [sys.executable, "-m", "mpi4jax_amd.run", "-n", "3",
+         "examples/master_worker.py", "--items", "12"],
+        capture_output=True, text=True, timeout=300, env=ENV, cwd=REPO,
+    )
+    assert res.returncode == 0, res.stderr + res.stdout
+    assert "master: 12/12 items verified OK" in res.stdout

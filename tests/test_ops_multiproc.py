@@ -308,3 +308,37 @@ def test_bitwise_reductions():
 
 def test_bitwise_reductions_world4():
     run_multiproc(_bitwise, 4)
+
+
+def _tagged_wildcards(rank, ws):
+    """Tagged send + wildcard recv on the CPU backend (the gloo-tag
+    implementation hung here and echoed the requested tag into Status —
+    found by the master/worker example; fixed by riding the envelope
+    plane).  Matching must be MPI's earliest-send rule with the ACTUAL
+    tag in Status."""
+    if rank == 0:
+        m.send(torch.full((2,), 1.0), dest=1, tag=5)
+        m.send(torch.full((3,), 2.0), dest=1, tag=9)
+        m.send(torch.full((4,), 3.0), dest=1, tag=5)
+    elif rank == 1:
+        st = m.Status()
+        y = m.recv(torch.empty(3), source=0, tag=9, status=st)
+        assert torch.equal(y, torch.full((3,), 2.0))
+        assert (st.source, st.tag) == (0, 9)
+        st = m.Status()
+        y = m.recv(torch.empty(2), source=m.ANY_SOURCE, tag=m.ANY_TAG,
+                   status=st)
+        assert torch.equal(y, torch.full((2,), 1.0)), "earliest send wins"
+        assert (st.source, st.tag) == (0, 5)
+        st = m.Status()
+        y = m.recv(torch.empty(4), source=0, tag=m.ANY_TAG, status=st)
+        assert torch.equal(y, torch.full((4,), 3.0))
+        assert st.tag == 5
+    # a wildcard SEND tag must raise on every backend
+    import pytest as _pytest
+    with _pytest.raises(ValueError, match="sendable"):
+        m.send(torch.ones(1), dest=(rank + 1) % ws, tag=m.ANY_TAG)
+
+
+def test_tagged_wildcard_matching_world2():
+    run_multiproc(_tagged_wildcards, 2)
