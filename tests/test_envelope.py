@@ -136,3 +136,79 @@ def _proto_worker(rank, ws):
 
 def test_envelope_protocol_world2_gloo():
     run_multiproc(_proto_worker, 2)
+
+
+# ------------------------------------------------- property-based model
+try:
+    from hypothesis import given, settings, strategies as st
+
+    HAVE_HYPOTHESIS = True
+except ImportError:  # pragma: no cover
+    HAVE_HYPOTHESIS = False
+
+
+if HAVE_HYPOTHESIS:
+    @settings(max_examples=200, deadline=None)
+    @given(st.data())
+    def test_claim_matches_mpi_model(data):
+        """EnvelopeBox.claim vs MPI's guarantees: for any interleaving
+        of arrivals and (source, tag) queries, the claimed message must
+        match the query, be that SOURCE's earliest unconsumed matching
+        message (per-pair non-overtaking — MPI does not order across
+        sources), and data drains per source must follow send order;
+        nothing may be left once everything is consumed."""
+        n_src = data.draw(st.integers(1, 3))
+        sends = data.draw(st.lists(
+            st.tuples(st.integers(0, n_src - 1), st.integers(0, 3),
+                      st.integers(1, 5)),
+            min_size=1, max_size=12))
+        box = _box(rank=3, size=4)
+        # sent[k] = (src, tag, nbytes); serial k encodes payload
+        pending = list(enumerate(sends))  # model: unconsumed, send order
+        arrived = 0
+        drains = {s: [] for s in range(n_src)}
+
+        def recv_bytes(src, nbytes):
+            drains[src].append(nbytes)
+            return torch.full((nbytes,), 0, dtype=torch.uint8)
+
+        while pending:
+            # deliver a random number of envelopes ahead of the query
+            k = data.draw(st.integers(0, len(sends) - arrived))
+            for _ in range(k):
+                s, t, nb = sends[arrived]
+                box.queue.append([s, t, nb])
+                arrived += 1
+            # query a message that is guaranteed claimable (its envelope
+            # either arrived or claim would block in _recv_one)
+            claimable = [e for e in pending if e[0] < arrived]
+            if not claimable:
+                s, t, nb = sends[arrived]
+                box.queue.append([s, t, nb])
+                arrived += 1
+                claimable = [e for e in pending if e[0] < arrived]
+            serial, (esrc, etag, enb) = data.draw(st.sampled_from(claimable))
+            use_any_src = data.draw(st.booleans())
+            use_any_tag = data.draw(st.booleans())
+            q_src = ANY_SOURCE if use_any_src else esrc
+            q_tag = ANY_TAG if use_any_tag else etag
+            got_s, got_t, _ = box.claim(q_src, q_tag, recv_bytes)
+            # 1. the claim matches the query
+            assert q_src in (ANY_SOURCE, got_s)
+            assert q_tag in (ANY_TAG, got_t)
+            # 2. non-overtaking: it is the claimed SOURCE's earliest
+            #    unconsumed matching message
+            exp = next(e for e in pending if e[1][0] == got_s
+                       and (q_tag in (ANY_TAG, e[1][1])))
+            assert got_t == exp[1][1], (sends, exp, (got_s, got_t))
+            pending.remove(exp)
+        assert box.queue == []
+        # every stash entry must eventually have been claimed
+        assert box.stash == []
+        # drained messages follow send order per source (a subsequence
+        # of that source's nbytes sequence, in order)
+        for s in range(n_src):
+            sent_nb = [nb for (src, _t, nb) in sends if src == s]
+            it = iter(sent_nb)
+            assert all(any(nb == x for x in it) for nb in drains[s]), (
+                sends, s, drains[s])
