@@ -334,6 +334,13 @@ def _tagged_wildcards(rank, ws):
         y = m.recv(torch.empty(4), source=0, tag=m.ANY_TAG, status=st)
         assert torch.equal(y, torch.full((4,), 3.0))
         assert st.tag == 5
+    # zero-size messages carry their envelope like any other
+    if rank == 0:
+        m.send(torch.empty(0), dest=1, tag=3)
+    elif rank == 1:
+        st = m.Status()
+        y = m.recv(torch.empty(0), source=0, tag=m.ANY_TAG, status=st)
+        assert y.numel() == 0 and st.tag == 3 and st.count == 0
     # a wildcard SEND tag must raise on every backend
     import pytest as _pytest
     with _pytest.raises(ValueError, match="sendable"):
