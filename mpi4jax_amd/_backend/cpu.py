@@ -44,13 +44,32 @@ def _isend_data(comm, wire, dest):
     pl[:] = [(ww, b) for ww, b in pl if not ww.is_completed()]
 
 
-def drain_pending():
-    """Wait for all in-flight data sends (MPI_Finalize semantics: blocks
-    until every send has been matched and delivered)."""
+def drain_pending(timeout_s=60.0):
+    """Wait for all in-flight data sends (MPI_Finalize semantics: every
+    send must be matched and delivered).  Bounded: an UNMATCHED send at
+    finalize is a bug in the program — after ``timeout_s`` this warns
+    loudly and abandons the sends instead of hanging interpreter exit
+    (MPI implementations typically abort here)."""
+    import time
+    import warnings
+
+    deadline = time.monotonic() + timeout_s
+    leftover = 0
     for pl in _PENDING.values():
         for w, _ in pl:
-            w.wait()
+            while not w.is_completed():
+                if time.monotonic() > deadline:
+                    leftover += 1
+                    break
+                time.sleep(0.005)
         pl.clear()
+    if leftover:
+        warnings.warn(
+            f"mpi4jax_amd.finalize: {leftover} point-to-point send(s) "
+            "were never matched by a receive — abandoned after "
+            f"{timeout_s:.0f}s (check your send/recv pairing)",
+            stacklevel=2,
+        )
 
 
 def _env_recv_into(out, source, tag, comm, status):
