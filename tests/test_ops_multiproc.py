@@ -349,3 +349,29 @@ def _tagged_wildcards(rank, ws):
 
 def test_tagged_wildcard_matching_world2():
     run_multiproc(_tagged_wildcards, 2)
+
+
+def _split_tagged(rank, ws):
+    """Tagged wildcards on Split sub-communicators: each comm has its own
+    envelope plane (per-comm gloo group + box), so identical tags on
+    sibling comms never cross."""
+    world = m.get_world()
+    sub = world.Split(color=rank // 2)  # {0,1} and {2,3}
+    peer = 1 - sub.rank
+    if sub.rank == 0:
+        m.send(torch.full((2,), float(rank)), dest=peer, tag=7, comm=sub)
+        st = m.Status()
+        y = m.recv(torch.empty(2), source=m.ANY_SOURCE, tag=m.ANY_TAG,
+                   comm=sub, status=st)
+    else:
+        st = m.Status()
+        y = m.recv(torch.empty(2), source=m.ANY_SOURCE, tag=m.ANY_TAG,
+                   comm=sub, status=st)
+        m.send(torch.full((2,), float(rank)), dest=peer, tag=7, comm=sub)
+    exp = float((rank // 2) * 2 + peer)
+    assert y[0].item() == exp, (rank, y[0].item(), exp)
+    assert st.tag == 7 and st.source == peer
+
+
+def test_split_comm_tagged_isolation_world4():
+    run_multiproc(_split_tagged, 4)
