@@ -412,6 +412,13 @@ def recv(template, source, tag, comm, status):
 
 
 def sendrecv(sendbuf, recvbuf, source, dest, sendtag, recvtag, comm, status):
+    if source == ANY_SOURCE:
+        raise ValueError(
+            "sendrecv: ANY_SOURCE is not supported on the RCCL backend "
+            "(the grouped send+recv pair needs an explicit peer; use "
+            "separate send/recv with MPI4JAX_AMD_GPU_ENVELOPE=1 for "
+            "wildcard receives)"
+        )
     with debug_timer("Sendrecv", comm.rank, f"src {source} dst {dest}"):
         out = torch.empty(tuple(recvbuf.shape), dtype=recvbuf.dtype,
                           device=recvbuf.device)

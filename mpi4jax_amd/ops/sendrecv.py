@@ -11,7 +11,7 @@ import torch
 
 from ..utils.tokens import NOTSET, raise_if_token_is_set
 from ..utils.validation import enforce_types
-from ..utils.status import ANY_TAG, Status
+from ..utils.status import ANY_SOURCE, ANY_TAG, Status
 from ._common import prepare, as_tensor
 from ..utils.dtypes import check_dtype
 
@@ -40,6 +40,11 @@ class _Sendrecv(torch.autograd.Function):
         # which is not a sendable tag — so the backward pair always uses
         # (sendtag=0, recvtag=ANY_TAG): both sides agree by construction.
         source, dest, _sendtag, _recvtag, comm, backend = ctx.meta
+        if source == ANY_SOURCE:
+            raise RuntimeError(
+                "grad through sendrecv(source=ANY_SOURCE) is undefined — "
+                "the reversed edge is unknown; pass an explicit source"
+            )
         template = grad.new_empty(ctx.send_shape)
         grad_send = backend.sendrecv(
             grad.contiguous(), template, dest, source, 0, ANY_TAG,
@@ -91,6 +96,8 @@ def sendrecv(sendbuf, recvbuf, source, dest, *, sendtag=0, recvtag=ANY_TAG,
             f"({recvbuf.device}) must live on the same device"
         )
     for name, r in (("source", source), ("dest", dest)):
+        if name == "source" and r == ANY_SOURCE:
+            continue  # MPI_Sendrecv accepts ANY_SOURCE for the receive
         if not 0 <= r < comm.size:
             raise ValueError(
                 f"sendrecv: invalid {name} {r} for comm size {comm.size}"

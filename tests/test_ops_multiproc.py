@@ -341,10 +341,18 @@ def _tagged_wildcards(rank, ws):
         st = m.Status()
         y = m.recv(torch.empty(0), source=0, tag=m.ANY_TAG, status=st)
         assert y.numel() == 0 and st.tag == 3 and st.count == 0
+    # sendrecv accepts ANY_SOURCE for its receive half (MPI_Sendrecv
+    # semantics) on the CPU envelope plane
+    peer = (rank + 1) % ws
+    st = m.Status()
+    y = m.sendrecv(torch.full((2,), float(rank)), torch.empty(2),
+                   source=m.ANY_SOURCE, dest=peer, status=st)
+    assert y[0].item() == float((rank - 1) % ws)
+    assert st.source == (rank - 1) % ws
     # a wildcard SEND tag must raise on every backend
     import pytest as _pytest
     with _pytest.raises(ValueError, match="sendable"):
-        m.send(torch.ones(1), dest=(rank + 1) % ws, tag=m.ANY_TAG)
+        m.send(torch.ones(1), dest=peer, tag=m.ANY_TAG)
 
 
 def test_tagged_wildcard_matching_world2():
