@@ -291,6 +291,28 @@ def main():
     # warmup (untimed) + hipGraph capture of a multistep
     for _ in range(args.warmup):
         state = sw.step(state)
+
+    # all-rank finiteness gate after warmup: if the fused step ever
+    # misbehaves on a topology this pool could not test (ROADMAP #1),
+    # fall back LOUDLY to the two-kernel pipeline rather than dying —
+    # the fallback is printed and carried in the result JSON.
+    fused_fallback = False
+    if sw.fused:
+        ok = torch.isfinite(state.h).all().to(torch.float32).cpu()
+        ok = m.allreduce(ok, m.MIN, comm=comm).item()
+        if ok < 1.0:
+            import os as _os
+
+            print("# fused step went non-finite after warmup; "
+                  "falling back to MPI4JAX_AMD_SW_NOFUSE=1", flush=True)
+            _os.environ["MPI4JAX_AMD_SW_NOFUSE"] = "1"
+            fused_fallback = True
+            sw = ShallowWater(nx=args.nx, ny=args.ny, comm=comm,
+                              dims=dims, device=device,
+                              dtype=torch.float32)
+            state = sw.step(sw.initial_conditions(), first_step=True)
+            for _ in range(args.warmup):
+                state = sw.step(state)
     # steps per captured graph: bigger graphs amortize replay-launch
     # overhead (87k-step soak measures 0.0911 ms/step at spc=500 vs
     # 0.101 at spc=10); capture+validation cost stays untimed either way
@@ -364,6 +386,7 @@ def main():
                 "sec_per_model_day": round(sec_per_model_day, 2),
                 "parallelism": f"domain-decomposition {dims[0]}x{dims[1]}",
                 "fused_kernels": sw.fused,
+                "fused_fallback": fused_fallback,
                 **comm_bench,
             },
         }
