@@ -1,13 +1,19 @@
-"""Bootstrap-plane message envelopes for the GPU p2p path.
+"""Bootstrap-plane message envelopes: MPI matching for order-matched
+data planes.
 
-RCCL has no message envelope: remote send/recv match purely by enqueue
-order per (sender, receiver) pair.  That forbids MPI-style tag matching
-and ``recv(ANY_SOURCE)`` on the GPU — the two documented p2p parity gaps
-vs the reference (``mpi_ops_common.h:354-367`` transmits both through
-MPI).  With ``MPI4JAX_AMD_GPU_ENVELOPE=1`` every remote GPU send posts a
-tiny CPU envelope ``(tag, nbytes)`` over the communicator's gloo
-bootstrap group; ``recv`` consumes envelopes to decide WHICH RCCL message
-to take next:
+Neither RCCL nor a single gloo wire tag has a message envelope: remote
+send/recv match purely by enqueue order per (sender, receiver) pair.
+That forbids MPI-style tag matching and ``recv(ANY_SOURCE)`` — the p2p
+machinery the reference gets from MPI itself
+(``mpi_ops_common.h:354-367``).  This plane restores it for both
+backends: every remote send posts a tiny envelope ``(tag, nbytes)`` over
+the communicator's gloo bootstrap group, and ``recv`` consumes envelopes
+to decide WHICH data-plane message to take next.  The CPU backend rides
+it unconditionally (``_backend/cpu.py``); the RCCL backend behind
+``MPI4JAX_AMD_GPU_ENVELOPE=1`` (default off: the envelope costs ~10 µs
+of host latency per message, and order-matched default-tag p2p — every
+reference-style workload — needs none of it).  On the GPU path recv
+decides:
 
 * envelope matches in order → direct ``ncclRecv`` into the output;
 * a differently-tagged message is ahead in the pipe → it is
