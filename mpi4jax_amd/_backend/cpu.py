@@ -306,6 +306,11 @@ def recv(template, source, tag, comm, status):
             raise RuntimeError(
                 "recv from self with no matching buffered send"
             )
+        if source == ANY_SOURCE and comm.size == 1:
+            raise ValueError(
+                "recv: ANY_SOURCE with no remote peers (comm size 1) — "
+                "buffered self-sends need an explicit source=comm.rank"
+            )
         return _env_recv_into(out, source, tag, comm, status)
 
 

@@ -376,6 +376,11 @@ def recv(template, source, tag, comm, status):
             "plane (set MPI4JAX_AMD_GPU_ENVELOPE=1) — RCCL itself has no "
             "message envelope; alternatively pass an explicit source rank"
         )
+    if source == ANY_SOURCE and comm.size == 1:
+        raise ValueError(
+            "recv: ANY_SOURCE with no remote peers (comm size 1) — "
+            "buffered self-sends need an explicit source=comm.rank"
+        )
     with debug_timer("Recv", comm.rank, f"from {source}, tag {tag}"):
         out = torch.empty(tuple(template.shape), dtype=template.dtype,
                           device=template.device)
