@@ -696,9 +696,11 @@ def test_gpu_tag_hard_failure(monkeypatch):
 
     class FakeComm:  # pretends we are rank 1 so peer 0 is remote
         rank, size = 1, 2
+        gloo_group = None  # no bootstrap plane -> envelope unavailable
 
     x = torch.ones(4, device="cuda")
     monkeypatch.delenv("MPI4JAX_AMD_ALLOW_GPU_TAGS", raising=False)
+    monkeypatch.delenv("MPI4JAX_AMD_GPU_ENVELOPE", raising=False)
     with pytest.raises(ValueError, match="tag 7"):
         rccl.send(x, 0, 7, FakeComm())
     with pytest.raises(ValueError, match="tag 9"):
